@@ -1,0 +1,124 @@
+"""Benchmark contract for the driver.
+
+Measures the BASELINE.json north-star metric: images/sec (whole node) for
+CycleGAN horse2zebra-shaped training at 256x256, bf16 compute, 9-resblock
+generator, per-GPU batch 4, synthetic data + random-init weights.
+
+``images`` counts every image consumed by a train step: a step processes
+``global_batch`` horse images AND ``global_batch`` zebra images, so
+images/step = 2 * global_batch (pairs/sec = value / 2, also reported).
+
+Usage:  python bench.py [--gpus N] [--steps K] [--warmup W]
+Multi-GPU (driver):  python -m torch.distributed.run --nnodes=1
+    --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from cyclegan_amd.parallel import DistContext
+from cyclegan_amd.trainer import CycleGAN
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch_size", type=int, default=4, help="per-GPU batch")
+    ap.add_argument("--image_size", type=int, default=256)
+    ap.add_argument("--num_residual_blocks", type=int, default=9)
+    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--output_dir", default="/tmp/cyclegan_bench")
+    args = ap.parse_args()
+
+    ctx = DistContext()
+    on_gpu = ctx.device.type == "cuda"
+    if not on_gpu:
+        # plumbing-only CPU fallback (BASELINE config 1); the measured
+        # configuration requires the MI355X box.
+        args.image_size = 64
+        args.num_residual_blocks = 1
+        args.batch_size = 1
+        args.dtype = "fp32"
+
+    torch.manual_seed(1234)
+    args.global_batch_size = ctx.world_size * args.batch_size
+    args.compute_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    os.makedirs(args.output_dir, exist_ok=True)
+
+    gan = CycleGAN(args, ctx)
+
+    # synthetic input pool, generated on-device (no host pipeline in the
+    # timed region; data shape = the horse2zebra 256x256 config)
+    g = torch.Generator(device="cpu").manual_seed(4321 + ctx.rank)
+    pool = []
+    for _ in range(4):
+        x = torch.rand(args.batch_size, args.image_size, args.image_size, 3,
+                       generator=g) * 2 - 1
+        y = torch.rand(args.batch_size, args.image_size, args.image_size, 3,
+                       generator=g) * 2 - 1
+        pool.append((x.to(ctx.device, args.compute_dtype),
+                     y.to(ctx.device, args.compute_dtype)))
+
+    def sync():
+        ctx.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        gan.train_step(*pool[i % len(pool)])
+    sync()
+
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        gan.train_step(*pool[i % len(pool)])
+    if on_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    ctx.barrier()
+
+    # MAX elapsed over ranks -> whole-job throughput
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if ctx.distributed:
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    elapsed = t.item()
+
+    images_per_step = 2 * args.global_batch_size
+    value = images_per_step * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if ctx.is_main:
+        print(json.dumps({
+            "metric": "images/sec (whole node) horse2zebra 256×256",
+            "value": round(value, 3),
+            "unit": "images/s",
+            "n_gpus": ctx.world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": f"cyclegan-resnet{args.num_residual_blocks}-patchgan",
+                "global_batch": args.global_batch_size,
+                "per_gpu_batch": args.batch_size,
+                "image_size": args.image_size,
+                "images_per_step": images_per_step,
+                "pairs_per_sec": round(value / 2, 3),
+                "parallelism": f"dp{ctx.world_size}",
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

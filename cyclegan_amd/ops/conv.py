@@ -1,0 +1,210 @@
+"""NHWC convolution ops (implicit-GEMM MFMA kernels on GPU).
+
+Layouts (MI355X-first):
+- activations: ``[B, H, W, C]`` contiguous (channels innermost → coalesced
+  bf16x8 loads, C is the GEMM K/N dimension for the matrix cores).
+- weights: HWIO ``[kh, kw, Cin, Cout]`` contiguous — the implicit-GEMM B
+  operand ``[K = kh*kw*Cin, N = Cout]`` row-major, no transpose needed.
+- conv_transpose weights: HWIO ``[kh, kw, Cin, Cout]`` with Cin = source
+  channels (our own convention; checkpoints are native to this framework).
+
+Padding follows the TF 'SAME'/'VALID' conventions of the reference
+(pad_before = total//2 — /root/reference/main.py uses Keras Conv2D 'same');
+pads are explicit ``(pt, pb, pl, pr)`` everywhere. ``pad_mode='reflect'``
+implements ReflectionPadding2D folded into the conv
+(/root/reference/cyclegan/model.py:14-33).
+
+Activation epilogues (none/relu/lrelu/tanh) are fused into the conv kernel
+— on 8 TB/s HBM the win is not re-reading the output tensor.
+
+Autograd: the Function receives the fp32 master weight; bf16 compute copies
+come from ops.shadow. Weight grads are produced in fp32 (split-K
+accumulation) and flow straight into the fp32 master's grad bucket.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from . import backend
+from .shadow import compute_weight
+
+ACT_NONE, ACT_RELU, ACT_LRELU, ACT_TANH = 0, 1, 2, 3
+
+_ACT = {None: ACT_NONE, "none": ACT_NONE, "relu": ACT_RELU, "lrelu": ACT_LRELU, "tanh": ACT_TANH}
+
+
+def same_pads(in_h: int, in_w: int, kh: int, kw: int, stride: int) -> Tuple[int, int, int, int]:
+    """TF 'SAME' pads (pad_before = total//2, extra goes after)."""
+    out_h = -(-in_h // stride)
+    out_w = -(-in_w // stride)
+    ph = max(0, (out_h - 1) * stride + kh - in_h)
+    pw = max(0, (out_w - 1) * stride + kw - in_w)
+    return ph // 2, ph - ph // 2, pw // 2, pw - pw // 2
+
+
+def _apply_act(y: torch.Tensor, act: int, slope: float) -> torch.Tensor:
+    if act == ACT_RELU:
+        return torch.relu(y)
+    if act == ACT_LRELU:
+        return F.leaky_relu(y, slope)
+    if act == ACT_TANH:
+        return torch.tanh(y)
+    return y
+
+
+def act_bwd_from_output(dy: torch.Tensor, y: torch.Tensor, act: int, slope: float) -> torch.Tensor:
+    """Backward of the fused activation, computed from the *output* y."""
+    if act == ACT_RELU:
+        return dy * (y > 0).to(dy.dtype)
+    if act == ACT_LRELU:
+        # slope > 0 keeps sign(pre-act) == sign(post-act)
+        return torch.where(y > 0, dy, dy * slope)
+    if act == ACT_TANH:
+        return dy * (1 - y.float() * y.float()).to(dy.dtype)
+    return dy
+
+
+def _conv_ref(x, w, bias, stride, pads, pad_mode):
+    """Differentiable torch reference (CPU path / GPU oracle)."""
+    pt, pb, pl, pr = pads
+    xn = x.permute(0, 3, 1, 2)
+    if pt or pb or pl or pr:
+        mode = "reflect" if pad_mode == "reflect" else "constant"
+        xn = F.pad(xn, (pl, pr, pt, pb), mode=mode)
+    wn = w.permute(3, 2, 0, 1)
+    y = F.conv2d(xn.contiguous(), wn.contiguous(), bias, stride=stride)
+    return y.permute(0, 2, 3, 1)
+
+
+def _convt_ref(x, w, bias, stride, pt, pl, out_h, out_w):
+    """Transpose conv = adjoint of the TF-'SAME' strided conv (gather form:
+    out[i] += in[o]*w[k] where i = s*o + k - pt)."""
+    xn = x.permute(0, 3, 1, 2)
+    wn = w.permute(2, 3, 0, 1)  # (Cin, Cout, kh, kw) for conv_transpose
+    y = F.conv_transpose2d(xn.contiguous(), wn.contiguous(), bias, stride=stride)
+    y = y[:, :, pt : pt + out_h, pl : pl + out_w]
+    return y.permute(0, 2, 3, 1)
+
+
+class _ConvFn(torch.autograd.Function):
+    """HIP implicit-GEMM conv with fused pad + bias + activation."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, pads, reflect, act, slope):
+        ext = backend.ext()
+        wc = compute_weight(w, x)
+        bc = compute_weight(bias, x) if bias is not None else None
+        y = ext.conv2d_fwd(x, wc, bc, stride, *pads, reflect, act, slope)
+        ctx.save_for_backward(x, w, y)
+        ctx.conf = (stride, pads, reflect, act, slope, bias is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y = ctx.saved_tensors
+        stride, pads, reflect, act, slope, has_bias = ctx.conf
+        ext = backend.ext()
+        dy = dy.contiguous()
+        if act != ACT_NONE:
+            dy = ext.act_bwd(dy, y, act, slope)
+        wc = compute_weight(w, x)
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.conv2d_dgrad(dy, wc, x.shape[1], x.shape[2], stride, *pads, reflect)
+        if ctx.needs_input_grad[1]:
+            dw = ext.conv2d_wgrad(x, dy, w.shape[0], w.shape[1], stride, *pads, reflect)
+            if dw.dtype != w.dtype:
+                dw = dw.to(w.dtype)
+        if has_bias and ctx.needs_input_grad[2]:
+            db = dy.float().sum(dim=(0, 1, 2))
+        return dx, dw, db, None, None, None, None, None
+
+
+class _ConvTFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, pt, pl, out_h, out_w, act, slope):
+        ext = backend.ext()
+        wc = compute_weight(w, x)
+        bc = compute_weight(bias, x) if bias is not None else None
+        y = ext.convt2d_fwd(x, wc, bc, stride, pt, pl, out_h, out_w, act, slope)
+        ctx.save_for_backward(x, w, y)
+        ctx.conf = (stride, pt, pl, act, slope, bias is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y = ctx.saved_tensors
+        stride, pt, pl, act, slope, has_bias = ctx.conf
+        ext = backend.ext()
+        dy = dy.contiguous()
+        if act != ACT_NONE:
+            dy = ext.act_bwd(dy, y, act, slope)
+        wc = compute_weight(w, x)
+        dx = dw = db = None
+        # adjoint of the gather is the forward strided conv
+        if ctx.needs_input_grad[0]:
+            dx = ext.convt2d_dgrad(dy, wc, x.shape[1], x.shape[2], stride, pt, pl)
+        if ctx.needs_input_grad[1]:
+            dw = ext.convt2d_wgrad(x, dy, w.shape[0], w.shape[1], stride, pt, pl)
+            if dw.dtype != w.dtype:
+                dw = dw.to(w.dtype)
+        if has_bias and ctx.needs_input_grad[2]:
+            db = dy.float().sum(dim=(0, 1, 2))
+        return dx, dw, db, None, None, None, None, None, None, None
+
+
+def conv2d(
+    x: torch.Tensor,
+    w: torch.Tensor,
+    bias: Optional[torch.Tensor] = None,
+    stride: int = 1,
+    padding: str | Tuple[int, int, int, int] = "valid",
+    pad_mode: str = "zeros",
+    act: Optional[str] = None,
+    slope: float = 0.2,
+) -> torch.Tensor:
+    """NHWC conv2d. ``padding`` is 'valid', 'same' (TF semantics) or explicit
+    (pt, pb, pl, pr). ``pad_mode`` 'zeros'|'reflect'. ``act`` fused epilogue."""
+    kh, kw = w.shape[0], w.shape[1]
+    if padding == "valid":
+        pads = (0, 0, 0, 0)
+    elif padding == "same":
+        pads = same_pads(x.shape[1], x.shape[2], kh, kw, stride)
+    else:
+        pads = tuple(padding)
+    a = _ACT[act]
+    if backend.use_hip(x, w):
+        return _ConvFn.apply(x, w, bias, stride, pads, pad_mode == "reflect", a, slope)
+    wc = w if w.dtype == x.dtype else w.to(x.dtype)
+    bc = bias if (bias is None or bias.dtype == x.dtype) else bias.to(x.dtype)
+    y = _conv_ref(x, wc, bc, stride, pads, pad_mode)
+    return _apply_act(y, a, slope)
+
+
+def conv_transpose2d(
+    x: torch.Tensor,
+    w: torch.Tensor,
+    bias: Optional[torch.Tensor] = None,
+    stride: int = 2,
+    act: Optional[str] = None,
+    slope: float = 0.2,
+) -> torch.Tensor:
+    """NHWC transpose conv with TF-'SAME' semantics: out = in*stride.
+
+    Defined as the adjoint of conv2d(..., stride, 'same') mapping
+    (in*stride) -> in, i.e. out[i] += in[o] * w[k] with i = s*o + k - pt.
+    """
+    kh, kw = w.shape[0], w.shape[1]
+    out_h, out_w = x.shape[1] * stride, x.shape[2] * stride
+    pt, _, pl, _ = same_pads(out_h, out_w, kh, kw, stride)
+    a = _ACT[act]
+    if backend.use_hip(x, w):
+        return _ConvTFn.apply(x, w, bias, stride, pt, pl, out_h, out_w, a, slope)
+    wc = w if w.dtype == x.dtype else w.to(x.dtype)
+    bc = bias if (bias is None or bias.dtype == x.dtype) else bias.to(x.dtype)
+    y = _convt_ref(x, wc, bc, stride, pt, pl, out_h, out_w)
+    return _apply_act(y, a, slope)

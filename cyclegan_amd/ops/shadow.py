@@ -1,0 +1,38 @@
+"""bf16 shadow-weight cache.
+
+Master parameters live in fp32 (optimizer numerics, exact DP all-reduce);
+compute on MI355X runs in bf16 on the MFMA matrix cores. Instead of
+recasting per call (the generator is invoked 3x per train step,
+/root/reference/main.py:207-262), each master tensor gets a cached bf16
+copy invalidated by the tensor's in-place version counter — the fused Adam
+step bumps ``_version``, so shadows refresh exactly once per optimizer step.
+"""
+
+from __future__ import annotations
+
+import torch
+from torch.utils.weak import WeakTensorKeyDictionary
+
+_cache: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
+
+
+def bf16_shadow(t: torch.Tensor) -> torch.Tensor:
+    """Return a cached bf16 copy of ``t`` (refreshed when t changes in-place)."""
+    if t.dtype == torch.bfloat16:
+        return t
+    ent = _cache.get(t)
+    ver = t._version
+    if ent is not None and ent[0] == ver:
+        return ent[1]
+    s = t.detach().to(torch.bfloat16)
+    _cache[t] = (ver, s)
+    return s
+
+
+def compute_weight(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
+    """Weight in the compute dtype of activation ``like`` (no autograd edge)."""
+    if like.dtype == torch.bfloat16 and w.dtype != torch.bfloat16:
+        return bf16_shadow(w)
+    if w.dtype != like.dtype:
+        return w.detach().to(like.dtype)
+    return w.detach()

@@ -1,0 +1,60 @@
+"""Flat parameter/gradient storage per model.
+
+Every parameter of a model becomes a view into one contiguous fp32 buffer,
+and its ``.grad`` a view into a matching flat grad buffer. Consequences:
+
+- gradient all-reduce is ONE RCCL call on the flat grad buffer per model
+  (the reference's 4 per-optimizer all-reduce groups S1-S4),
+  sized for xGMI (45.5 MB / 11 MB fp32) instead of 71/13 small messages;
+- Adam is one fused kernel over the flat pair (ops.adam);
+- bf16 shadow invalidation is automatic: views share the base buffer's
+  version counter, so an in-place flat update refreshes every shadow.
+"""
+
+from __future__ import annotations
+
+from typing import List
+
+import torch
+
+
+class FlatParamGroup:
+    def __init__(self, module: torch.nn.Module):
+        params: List[torch.nn.Parameter] = [p for p in module.parameters() if p.requires_grad]
+        assert all(p.dtype == torch.float32 for p in params), "master params must be fp32"
+        n = sum(p.numel() for p in params)
+        dev = params[0].device
+        self.flat_param = torch.empty(n, dtype=torch.float32, device=dev)
+        self.flat_grad = torch.zeros(n, dtype=torch.float32, device=dev)
+        self.params = params
+        self._offsets = []
+        off = 0
+        with torch.no_grad():
+            for p in params:
+                k = p.numel()
+                self.flat_param[off:off + k].copy_(p.detach().reshape(-1))
+                p.data = self.flat_param[off:off + k].view(p.shape)
+                self._offsets.append((off, k))
+                off += k
+        self.attach_grads()
+
+    def attach_grads(self):
+        """(Re)point every param.grad at its flat-grad view."""
+        for p, (off, k) in zip(self.params, self._offsets):
+            p.grad = self.flat_grad[off:off + k].view(p.shape)
+
+    def zero_grad(self):
+        self.flat_grad.zero_()
+        # autograd accumulates in-place into existing .grad views; re-attach
+        # defensively in case an engine path replaced one.
+        self.attach_grads()
+
+    def check_views(self) -> bool:
+        """True iff every param/grad still aliases the flat buffers."""
+        pp = self.flat_param.data_ptr()
+        gp = self.flat_grad.data_ptr()
+        ok = True
+        for p in self.params:
+            ok &= p.data.data_ptr() >= pp and p.data.data_ptr() < pp + self.flat_param.numel() * 4
+            ok &= p.grad is not None and gp <= p.grad.data_ptr() < gp + self.flat_grad.numel() * 4
+        return ok

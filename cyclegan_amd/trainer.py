@@ -1,0 +1,222 @@
+"""CycleGAN trainer.
+
+Replicates the reference trainer semantics (/root/reference/main.py:106-329)
+on the MI355X execution model:
+
+- 4 models (G: X->Y, F: Y->X, discriminators X, Y), 4 TF-style Adam
+  optimizers (2e-4, beta1=0.5, beta2=0.9, eps=1e-7) over flat fp32 buffers;
+- losses pre-scaled by 1/global_batch_size so SUM all-reduce over DP
+  replicas is the exact global mean (main.py:172-174);
+- per-step: one shared forward mega-graph; 4 gradient passes restricted to
+  each model's variables (``torch.autograd.backward(..., inputs=group)`` ==
+  TF ``optimizer.minimize(var_list=...)``), each followed immediately by an
+  async RCCL all-reduce of that group's flat gradient, overlapping the next
+  backward (reference runs the 4 groups serially, main.py:249-260);
+- G's adversarial gradient flows through the frozen discriminator (the
+  ``inputs=`` restriction limits accumulation, not the gradient path);
+  the discriminator update uses re-discriminated detached fakes, matching
+  the reference's recomputed X(fake_x)/Y(fake_y) calls (main.py:239-245);
+- compute dtype bf16 on GPU (fp32 masters), fp32 on CPU.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Dict, Optional
+
+import torch
+
+from .models import Generator, Discriminator
+from .ops import MAE, MSE, MSE_const
+from .ops.adam import FusedAdam
+from .parallel import DistContext, GradSync, FlatParamGroup
+
+
+class CycleGAN:
+    LAMBDA_CYCLE = 10.0
+    LAMBDA_IDENTITY = 0.5 * 10.0
+
+    def __init__(self, args, ctx: DistContext):
+        self.ctx = ctx
+        self.device = ctx.device
+        self.global_batch_size = args.global_batch_size
+        self.compute_dtype = getattr(args, "compute_dtype", None) or (
+            torch.bfloat16 if self.device.type == "cuda" else torch.float32)
+
+        self.checkpoint_dir = os.path.join(args.output_dir, "checkpoints")
+        if ctx.is_main:
+            os.makedirs(self.checkpoint_dir, exist_ok=True)
+        self.checkpoint_path = os.path.join(self.checkpoint_dir, "checkpoint.pt")
+
+        nrb = getattr(args, "num_residual_blocks", 9)
+        self.G = Generator(num_residual_blocks=nrb).to(self.device)
+        self.F = Generator(num_residual_blocks=nrb).to(self.device)
+        self.X = Discriminator().to(self.device)
+        self.Y = Discriminator().to(self.device)
+
+        # mirror rank-0 init to every replica (reference S7 broadcast)
+        for m in (self.G, self.F, self.X, self.Y):
+            ctx.broadcast_module(m)
+
+        self.groups = {name: FlatParamGroup(m) for name, m in
+                       (("G", self.G), ("F", self.F), ("X", self.X), ("Y", self.Y))}
+        self.optimizers = {name: FusedAdam(g.flat_param, g.flat_grad)
+                           for name, g in self.groups.items()}
+        self.sync = GradSync(ctx)
+
+    # ---- loss functions (reference main.py:172-195) ----
+
+    def reduce_mean(self, per_sample: torch.Tensor) -> torch.Tensor:
+        return per_sample.sum() / self.global_batch_size
+
+    def generator_loss(self, discriminate_fake):
+        return self.reduce_mean(MSE_const(discriminate_fake, 1.0))
+
+    def cycle_loss(self, real, cycled):
+        return self.LAMBDA_CYCLE * self.reduce_mean(MAE(real, cycled))
+
+    def identity_loss(self, real, same):
+        return self.LAMBDA_IDENTITY * self.reduce_mean(MAE(real, same))
+
+    def discriminator_loss(self, discriminate_real, discriminate_fake):
+        real = MSE_const(discriminate_real, 1.0)
+        fake = MSE_const(discriminate_fake, 0.0)
+        return self.reduce_mean(0.5 * (real + fake))
+
+    # ---- steps ----
+
+    def _cast(self, t: torch.Tensor) -> torch.Tensor:
+        return t.to(self.device, self.compute_dtype, non_blocking=True)
+
+    def train_step(self, x, y) -> Dict[str, torch.Tensor]:
+        x, y = self._cast(x), self._cast(y)
+        for g in self.groups.values():
+            g.zero_grad()
+
+        fake_y = self.G(x)
+        fake_x = self.F(y)
+
+        discriminate_fake_x = self.X(fake_x)
+        discriminate_fake_y = self.Y(fake_y)
+
+        G_loss = self.generator_loss(discriminate_fake_y)
+        F_loss = self.generator_loss(discriminate_fake_x)
+        G_cycle_loss = self.cycle_loss(y, self.G(fake_x))
+        F_cycle_loss = self.cycle_loss(x, self.F(fake_y))
+        G_identity_loss = self.identity_loss(y, self.G(y))
+        F_identity_loss = self.identity_loss(x, self.F(x))
+        G_total = G_loss + G_cycle_loss + G_identity_loss
+        F_total = F_loss + F_cycle_loss + F_identity_loss
+
+        # discriminator pass on real + re-discriminated (detached) fakes
+        X_loss = self.discriminator_loss(self.X(x), self.X(fake_x.detach()))
+        Y_loss = self.discriminator_loss(self.Y(y), self.Y(fake_y.detach()))
+
+        torch.autograd.backward(G_total, inputs=self.groups["G"].params,
+                                retain_graph=True)
+        self.sync.launch(self.groups["G"].flat_grad)
+        torch.autograd.backward(F_total, inputs=self.groups["F"].params)
+        self.sync.launch(self.groups["F"].flat_grad)
+        torch.autograd.backward(X_loss, inputs=self.groups["X"].params)
+        self.sync.launch(self.groups["X"].flat_grad)
+        torch.autograd.backward(Y_loss, inputs=self.groups["Y"].params)
+        self.sync.launch(self.groups["Y"].flat_grad)
+        self.sync.wait_all()
+
+        for opt in self.optimizers.values():
+            opt.step()
+
+        return {
+            "loss_G/loss": G_loss.detach(), "loss_G/cycle": G_cycle_loss.detach(),
+            "loss_G/identity": G_identity_loss.detach(), "loss_G/total": G_total.detach(),
+            "loss_F/loss": F_loss.detach(), "loss_F/cycle": F_cycle_loss.detach(),
+            "loss_F/identity": F_identity_loss.detach(), "loss_F/total": F_total.detach(),
+            "loss_X/loss": X_loss.detach(), "loss_Y/loss": Y_loss.detach(),
+        }
+
+    @torch.no_grad()
+    def cycle_step(self, x, y, training: bool = False):
+        x, y = self._cast(x), self._cast(y)
+        fake_y = self.G(x)
+        cycle_x = self.F(fake_y)
+        fake_x = self.F(y)
+        cycle_y = self.G(fake_x)
+        return fake_x, fake_y, cycle_x, cycle_y
+
+    @torch.no_grad()
+    def test_step(self, x, y) -> Dict[str, torch.Tensor]:
+        x, y = self._cast(x), self._cast(y)
+        fake_x, fake_y, cycle_x, cycle_y = self.cycle_step(x, y)
+
+        discriminate_fake_x = self.X(fake_x)
+        discriminate_fake_y = self.Y(fake_y)
+        G_loss = self.generator_loss(discriminate_fake_y)
+        F_loss = self.generator_loss(discriminate_fake_x)
+        F_cycle_loss = self.cycle_loss(x, cycle_x)
+        G_cycle_loss = self.cycle_loss(y, cycle_y)
+        same_x = self.F(x)
+        same_y = self.G(y)
+        G_identity_loss = self.identity_loss(y, same_y)
+        F_identity_loss = self.identity_loss(x, same_x)
+        G_total = G_loss + G_cycle_loss + G_identity_loss
+        F_total = F_loss + F_cycle_loss + F_identity_loss
+        X_loss = self.discriminator_loss(self.X(x), discriminate_fake_x)
+        Y_loss = self.discriminator_loss(self.Y(y), discriminate_fake_y)
+
+        return {
+            "loss_G/loss": G_loss, "loss_G/cycle": G_cycle_loss,
+            "loss_G/identity": G_identity_loss, "loss_G/total": G_total,
+            "loss_F/loss": F_loss, "loss_F/cycle": F_cycle_loss,
+            "loss_F/identity": F_identity_loss, "loss_F/total": F_total,
+            "loss_X/loss": X_loss, "loss_Y/loss": Y_loss,
+            "error/MAE(X, F(G(X)))": self.reduce_mean(MAE(x, cycle_x)),
+            "error/MAE(Y, G(F(Y)))": self.reduce_mean(MAE(y, cycle_y)),
+            "error/MAE(X, F(X))": self.reduce_mean(MAE(x, same_x)),
+            "error/MAE(Y, G(Y))": self.reduce_mean(MAE(y, same_y)),
+        }
+
+    # ---- checkpoint (reference main.py:148-170: one overwriting prefix,
+    # all 4 models + 4 optimizer states, auto-resume if present) ----
+
+    def save_checkpoint(self):
+        if self.ctx.is_main:
+            state = {
+                "G": self.G.state_dict(), "F": self.F.state_dict(),
+                "X": self.X.state_dict(), "Y": self.Y.state_dict(),
+                "G_optimizer": self.optimizers["G"].state_dict(),
+                "F_optimizer": self.optimizers["F"].state_dict(),
+                "X_optimizer": self.optimizers["X"].state_dict(),
+                "Y_optimizer": self.optimizers["Y"].state_dict(),
+            }
+            tmp = self.checkpoint_path + ".tmp"
+            torch.save(state, tmp)
+            os.replace(tmp, self.checkpoint_path)
+            print(f"\nsaved checkpoint to {self.checkpoint_path}\n")
+        self.ctx.barrier()
+
+    def load_checkpoint(self) -> bool:
+        if not os.path.exists(self.checkpoint_path):
+            return False
+        state = torch.load(self.checkpoint_path, map_location=self.device,
+                           weights_only=True)
+        self.G.load_state_dict(state["G"])
+        self.F.load_state_dict(state["F"])
+        self.X.load_state_dict(state["X"])
+        self.Y.load_state_dict(state["Y"])
+        for name in ("G", "F", "X", "Y"):
+            self.optimizers[name].load_state_dict(state[f"{name}_optimizer"])
+        print(f"\nloaded checkpoint from {self.checkpoint_path}\n")
+        return True
+
+    # ---- epoch-level metric reduction (S5, deferred to epoch end) ----
+
+    def reduce_results(self, accumulated: Dict[str, list]) -> Dict[str, float]:
+        """Mean over steps, SUM all-reduce over replicas, to host floats."""
+        if not accumulated:
+            return {}
+        keys = list(accumulated.keys())
+        means = torch.stack([torch.stack([v.float() for v in accumulated[k]]).mean()
+                             for k in keys])
+        self.ctx.all_reduce_(means)
+        vals = means.tolist()
+        return dict(zip(keys, vals))

@@ -1,6 +1,6 @@
 """NHWC layer modules on top of cyclegan_amd.ops.
 
-Parameters are fp32 masters (HWIO for convs); compute dtype follows the
+Parameters are fp32 masters (OHWI for convs); compute dtype follows the
 activation dtype (bf16 on MI355X), with shadow casting handled inside the
 op layer.
 """
@@ -20,8 +20,8 @@ from ..ops.norm import EPS_DEFAULT
 def init_conv_weight_(w: torch.Tensor, kind: str = "normal002"):
     """Reference inits: RandomNormal(0, 0.02) for all convs except the
     generator head, which uses Keras' default glorot_uniform
-    (/root/reference/cyclegan/model.py:10-11,165-166)."""
-    kh, kw, cin, cout = w.shape
+    (/root/reference/cyclegan/model.py:10-11,165-166). w is OHWI."""
+    cout, kh, kw, cin = w.shape
     with torch.no_grad():
         if kind == "normal002":
             w.normal_(0.0, 0.02)
@@ -41,7 +41,7 @@ class ConvNHWC(nn.Module):
         super().__init__()
         self.stride, self.padding, self.pad_mode = stride, padding, pad_mode
         self.act, self.slope = act, slope
-        self.weight = nn.Parameter(torch.empty(kernel, kernel, cin, cout))
+        self.weight = nn.Parameter(torch.empty(cout, kernel, kernel, cin))
         init_conv_weight_(self.weight, init)
         if bias:
             self.bias = nn.Parameter(torch.zeros(cout))
@@ -61,7 +61,7 @@ class ConvTransposeNHWC(nn.Module):
                  init: str = "normal002"):
         super().__init__()
         self.stride, self.act = stride, act
-        self.weight = nn.Parameter(torch.empty(kernel, kernel, cin, cout))
+        self.weight = nn.Parameter(torch.empty(cout, kernel, kernel, cin))
         init_conv_weight_(self.weight, init)
         if bias:
             self.bias = nn.Parameter(torch.zeros(cout))

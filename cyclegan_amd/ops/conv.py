@@ -3,9 +3,12 @@
 Layouts (MI355X-first):
 - activations: ``[B, H, W, C]`` contiguous (channels innermost → coalesced
   bf16x8 loads, C is the GEMM K/N dimension for the matrix cores).
-- weights: HWIO ``[kh, kw, Cin, Cout]`` contiguous — the implicit-GEMM B
-  operand ``[K = kh*kw*Cin, N = Cout]`` row-major, no transpose needed.
-- conv_transpose weights: HWIO ``[kh, kw, Cin, Cout]`` with Cin = source
+- weights: OHWI ``[Cout, kh, kw, Cin]`` contiguous — the transposed
+  implicit-GEMM B operand ``B^T[N = Cout][K = kh*kw*Cin]`` row-major, so
+  weight staging is a coalesced row copy AND the per-lane MFMA B-fragment
+  (8 contiguous k at one n) is a single ds_read_b128. No transpose
+  anywhere on the forward path.
+- conv_transpose weights: OHWI ``[Cout, kh, kw, Cin]`` with Cin = source
   channels (our own convention; checkpoints are native to this framework).
 
 Padding follows the TF 'SAME'/'VALID' conventions of the reference
@@ -69,22 +72,22 @@ def act_bwd_from_output(dy: torch.Tensor, y: torch.Tensor, act: int, slope: floa
 
 
 def _conv_ref(x, w, bias, stride, pads, pad_mode):
-    """Differentiable torch reference (CPU path / GPU oracle)."""
+    """Differentiable torch reference (CPU path / GPU oracle); w is OHWI."""
     pt, pb, pl, pr = pads
     xn = x.permute(0, 3, 1, 2)
     if pt or pb or pl or pr:
         mode = "reflect" if pad_mode == "reflect" else "constant"
         xn = F.pad(xn, (pl, pr, pt, pb), mode=mode)
-    wn = w.permute(3, 2, 0, 1)
+    wn = w.permute(0, 3, 1, 2)  # OHWI -> OIHW
     y = F.conv2d(xn.contiguous(), wn.contiguous(), bias, stride=stride)
     return y.permute(0, 2, 3, 1)
 
 
 def _convt_ref(x, w, bias, stride, pt, pl, out_h, out_w):
     """Transpose conv = adjoint of the TF-'SAME' strided conv (gather form:
-    out[i] += in[o]*w[k] where i = s*o + k - pt)."""
+    out[i] += in[o]*w[k] where i = s*o + k - pt); w is OHWI."""
     xn = x.permute(0, 3, 1, 2)
-    wn = w.permute(2, 3, 0, 1)  # (Cin, Cout, kh, kw) for conv_transpose
+    wn = w.permute(3, 0, 1, 2)  # OHWI -> (Cin, Cout, kh, kw)
     y = F.conv_transpose2d(xn.contiguous(), wn.contiguous(), bias, stride=stride)
     y = y[:, :, pt : pt + out_h, pl : pl + out_w]
     return y.permute(0, 2, 3, 1)
@@ -116,7 +119,7 @@ class _ConvFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             dx = ext.conv2d_dgrad(dy, wc, x.shape[1], x.shape[2], stride, *pads, reflect)
         if ctx.needs_input_grad[1]:
-            dw = ext.conv2d_wgrad(x, dy, w.shape[0], w.shape[1], stride, *pads, reflect)
+            dw = ext.conv2d_wgrad(x, dy, w.shape[1], w.shape[2], stride, *pads, reflect)
             if dw.dtype != w.dtype:
                 dw = dw.to(w.dtype)
         if has_bias and ctx.needs_input_grad[2]:
@@ -149,7 +152,7 @@ class _ConvTFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             dx = ext.convt2d_dgrad(dy, wc, x.shape[1], x.shape[2], stride, pt, pl)
         if ctx.needs_input_grad[1]:
-            dw = ext.convt2d_wgrad(x, dy, w.shape[0], w.shape[1], stride, pt, pl)
+            dw = ext.convt2d_wgrad(x, dy, w.shape[1], w.shape[2], stride, pt, pl)
             if dw.dtype != w.dtype:
                 dw = dw.to(w.dtype)
         if has_bias and ctx.needs_input_grad[2]:
@@ -169,7 +172,7 @@ def conv2d(
 ) -> torch.Tensor:
     """NHWC conv2d. ``padding`` is 'valid', 'same' (TF semantics) or explicit
     (pt, pb, pl, pr). ``pad_mode`` 'zeros'|'reflect'. ``act`` fused epilogue."""
-    kh, kw = w.shape[0], w.shape[1]
+    kh, kw = w.shape[1], w.shape[2]
     if padding == "valid":
         pads = (0, 0, 0, 0)
     elif padding == "same":
@@ -198,7 +201,7 @@ def conv_transpose2d(
     Defined as the adjoint of conv2d(..., stride, 'same') mapping
     (in*stride) -> in, i.e. out[i] += in[o] * w[k] with i = s*o + k - pt.
     """
-    kh, kw = w.shape[0], w.shape[1]
+    kh, kw = w.shape[1], w.shape[2]
     out_h, out_w = x.shape[1] * stride, x.shape[2] * stride
     pt, _, pl, _ = same_pads(out_h, out_w, kh, kw, stride)
     a = _ACT[act]

@@ -13,17 +13,17 @@ from cyclegan_amd.ops.conv import act_bwd_from_output, ACT_RELU, ACT_LRELU, ACT_
 def test_conv2d_valid_matches_torch():
     torch.manual_seed(0)
     x = torch.randn(2, 10, 10, 3)
-    w = torch.randn(3, 3, 3, 8)
+    w = torch.randn(8, 3, 3, 3)  # OHWI
     y = ops.conv2d(x, w, padding="valid")
-    ref = F.conv2d(x.permute(0, 3, 1, 2), w.permute(3, 2, 0, 1))
+    ref = F.conv2d(x.permute(0, 3, 1, 2), w.permute(0, 3, 1, 2))
     assert torch.allclose(y, ref.permute(0, 2, 3, 1), atol=1e-5)
 
 
 def test_conv2d_same_stride2_output_shape():
     x = torch.randn(1, 256, 256, 4)
-    w = torch.randn(3, 3, 4, 8)
+    w = torch.randn(8, 3, 3, 4)
     assert ops.conv2d(x, w, stride=2, padding="same").shape == (1, 128, 128, 8)
-    w4 = torch.randn(4, 4, 4, 8)
+    w4 = torch.randn(8, 4, 4, 4)
     assert ops.conv2d(x, w4, stride=2, padding="same").shape == (1, 128, 128, 8)
     assert ops.conv2d(x, w4, stride=1, padding="same").shape == (1, 256, 256, 8)
 
@@ -31,7 +31,7 @@ def test_conv2d_same_stride2_output_shape():
 def test_conv2d_reflect_pad_equals_explicit_pad():
     torch.manual_seed(1)
     x = torch.randn(2, 8, 8, 4)
-    w = torch.randn(3, 3, 4, 4)
+    w = torch.randn(4, 3, 3, 4)
     y1 = ops.conv2d(x, w, padding=(1, 1, 1, 1), pad_mode="reflect")
     xp = ops.reflection_pad2d(x, (1, 1))
     y2 = ops.conv2d(xp, w, padding="valid")
@@ -43,11 +43,11 @@ def test_conv_transpose_is_adjoint_of_conv():
     torch.manual_seed(2)
     s = 2
     x = torch.randn(1, 16, 16, 6)
-    w = torch.randn(3, 3, 6, 4)  # conv: 6 -> 4 channels, 16 -> 8 spatial
+    w = torch.randn(4, 3, 3, 6)  # OHWI conv: 6 -> 4 channels, 16 -> 8 spatial
     y = torch.randn(1, 8, 8, 4)
     cx = ops.conv2d(x, w, stride=s, padding="same")
     # adjoint maps 8 -> 16 with the transposed channel order
-    wt = w.permute(0, 1, 3, 2).contiguous()  # [kh,kw,4,6]
+    wt = w.permute(3, 1, 2, 0).contiguous()  # OHWI with O=6, I=4
     aty = ops.conv_transpose2d(y, wt, stride=s)
     assert aty.shape == x.shape
     lhs = (cx * y).sum()
@@ -57,7 +57,7 @@ def test_conv_transpose_is_adjoint_of_conv():
 
 def test_conv_transpose_shape_tf_same():
     x = torch.randn(2, 64, 64, 8)
-    w = torch.randn(3, 3, 8, 4)
+    w = torch.randn(4, 3, 3, 8)
     assert ops.conv_transpose2d(x, w, stride=2).shape == (2, 128, 128, 4)
 
 
@@ -135,7 +135,7 @@ def test_act_bwd_from_output():
 def test_conv_backward_through_ref():
     torch.manual_seed(7)
     x = torch.randn(1, 8, 8, 3, requires_grad=True)
-    w = torch.randn(3, 3, 3, 4, requires_grad=True)
+    w = torch.randn(4, 3, 3, 3, requires_grad=True)
     y = ops.conv2d(x, w, stride=2, padding="same", act="relu")
     y.sum().backward()
     assert x.grad is not None and w.grad is not None
@@ -145,6 +145,6 @@ def test_conv_backward_through_ref():
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
 def test_conv_dtype_paths(dtype):
     x = torch.randn(1, 8, 8, 3).to(dtype)
-    w = torch.randn(3, 3, 3, 4)  # fp32 master
+    w = torch.randn(4, 3, 3, 3)  # fp32 master
     y = ops.conv2d(x, w, padding="same")
     assert y.dtype == dtype

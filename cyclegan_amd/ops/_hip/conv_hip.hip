@@ -1,0 +1,570 @@
+#include "hip/hip_runtime.h"
+// Implicit-GEMM NHWC convolutions on gfx950 MFMA (bf16 in, fp32 accumulate).
+//
+// Three kernels cover every conv op in the framework:
+//   conv_fwd_kernel   y[b,oh,ow,n] = act(Σ_{dk,ci} x[b,oh*s+dk-pt,ci]·w[n,dk,ci] + bias)
+//                     (also conv_transpose dgrad, via the channel-transposed weight)
+//   convt_fwd_kernel  y[b,i,j,n]   = act(Σ_{dk,ci; i=s*o+dk-pt} in[b,o,ci]·w[n,dk,ci] + bias)
+//                     (also conv2d dgrad: same gather, channel-transposed weight, no tap flip)
+//   wgrad_kernel      dw[n,dk,ci] += Σ_{b,o} x_patch[m,(dk,ci)]·dy[m,n]   (fp32, split-M atomics)
+//
+// GEMM view: M = B*OH*OW output pixels, N = Cout, K = KH*KW*Cin, with the
+// im2col A-tile gathered on the fly (reflection padding = a load-index
+// mirror, never materialized). Weights are OHWI so the B^T operand rows
+// (fixed n, k contiguous) are vector loads AND single ds_read_b128 MFMA
+// fragments.
+//
+// Tiling: 128x64 block tile, BK=64, 4 waves (2x2), 64x32 per wave,
+// mfma_f32_16x16x32_bf16, fragments: A row = lane&15, k = (lane>>4)*8..+8;
+// D col = lane&15, row = (lane>>4)*4 + reg (verified by tests/test_ops_gpu.py
+// via the mfma_probe binding and oracle comparisons).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace cyg {
+
+constexpr int BM = 128, BN = 64, BK = 64;
+constexpr int NTHREADS = 256;
+constexpr int LDK = BK + 8;  // +16B pad: conflict-free b128 column-of-rows reads
+
+struct ConvParams {
+  const short* __restrict__ x;   // [B,H,W,Cin] bf16 raw
+  const short* __restrict__ w;   // [Cout,KH,KW,Cin] bf16 raw
+  const short* __restrict__ bias;  // [Cout] bf16 or null
+  short* __restrict__ y;         // [B,OH,OW,Cout]
+  int B, H, W, Cin, OH, OW, Cout, KH, KW;
+  int stride, pt, pl;
+  int reflect;                   // conv_fwd only
+  int act;  float slope;
+  long M, KTOT;
+  int mtiles, ntiles;
+};
+
+// ---------------- shared GEMM core ----------------
+// As rows = m (output pixel), Bs rows = n (cout); both k-contiguous.
+
+template <bool IS_CONVT, bool ALIGNED>
+__global__ __launch_bounds__(NTHREADS) void conv_gemm_kernel(ConvParams p) {
+  __shared__ short As[BM][LDK];
+  __shared__ short Bs[BN][LDK];
+  __shared__ long rowxb[BM];    // batch base offset into x
+  __shared__ long rowyb[BM];    // output base offset (elements)
+  __shared__ int rowih[BM], rowiw[BM];  // gather base coords (see below)
+  __shared__ char rowok[BM];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int bid = blockIdx.x;
+  const int mt = bid % p.mtiles, nt = bid / p.mtiles;
+  const long m0 = (long)mt * BM;
+  const int n0 = nt * BN;
+
+  // ---- per-row precompute (constant across the K loop) ----
+  for (int r = tid; r < BM; r += NTHREADS) {
+    long m = m0 + r;
+    bool ok = m < p.M;
+    long mm = ok ? m : 0;
+    int ow = (int)(mm % p.OW);
+    int oh = (int)((mm / p.OW) % p.OH);
+    int b = (int)(mm / ((long)p.OW * p.OH));
+    rowok[r] = ok;
+    rowxb[r] = (long)b * p.H * p.W * p.Cin;
+    rowyb[r] = ((long)(b * p.OH + oh) * p.OW + ow) * p.Cout;
+    if (IS_CONVT) {
+      rowih[r] = oh + p.pt;   // output coord + pad (gather: o = (i+pt-dk)/s)
+      rowiw[r] = ow + p.pl;
+    } else {
+      rowih[r] = oh * p.stride - p.pt;
+      rowiw[r] = ow * p.stride - p.pl;
+    }
+  }
+  __syncthreads();
+
+  v4f acc[4][2] = {};
+
+  const int wr = wid >> 1, wc = wid & 1;
+  const int wm0 = wr * 64, wn0 = wc * 32;
+  const int fr = lane & 15;          // fragment row-in-16
+  const int fg = lane >> 4;          // k-group 0..3
+
+  for (long k0 = 0; k0 < p.KTOT; k0 += BK) {
+    // ---- stage A (im2col gather) ----
+    if (ALIGNED) {
+      #pragma unroll 2
+      for (int c = tid; c < BM * (BK / 8); c += NTHREADS) {
+        int row = c >> 3;
+        int kc = (c & 7) * 8;
+        long k = k0 + kc;
+        v8s val = {};
+        if (k < p.KTOT && rowok[row]) {
+          int tap = (int)(k / p.Cin);
+          int ci = (int)(k - (long)tap * p.Cin);
+          int dkh = tap / p.KW, dkw = tap - (tap / p.KW) * p.KW;
+          bool valid = true;
+          int ih, iw;
+          if (IS_CONVT) {
+            int nh = rowih[row] - dkh, nw = rowiw[row] - dkw;
+            valid = nh >= 0 && nw >= 0 && (nh % p.stride) == 0 &&
+                    (nw % p.stride) == 0;
+            ih = nh / p.stride; iw = nw / p.stride;
+            valid = valid && ih < p.H && iw < p.W;
+          } else {
+            ih = rowih[row] + dkh; iw = rowiw[row] + dkw;
+            if (p.reflect) {
+              ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
+            } else {
+              valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+            }
+          }
+          if (valid)
+            val = *(const v8s*)(p.x + rowxb[row] + ((long)ih * p.W + iw) * p.Cin + ci);
+        }
+        *(v8s*)&As[row][kc] = val;
+      }
+    } else {  // generic scalar path (Cin % 8 != 0, e.g. the RGB stem)
+      for (int c = tid; c < BM * BK; c += NTHREADS) {
+        int row = c >> 6;          // BK = 64
+        int kc = c & 63;
+        long k = k0 + kc;
+        short val = 0;
+        if (k < p.KTOT && rowok[row]) {
+          int tap = (int)(k / p.Cin);
+          int ci = (int)(k - (long)tap * p.Cin);
+          int dkh = tap / p.KW, dkw = tap - (tap / p.KW) * p.KW;
+          bool valid = true;
+          int ih, iw;
+          if (IS_CONVT) {
+            int nh = rowih[row] - dkh, nw = rowiw[row] - dkw;
+            valid = nh >= 0 && nw >= 0 && (nh % p.stride) == 0 &&
+                    (nw % p.stride) == 0;
+            ih = nh / p.stride; iw = nw / p.stride;
+            valid = valid && ih < p.H && iw < p.W;
+          } else {
+            ih = rowih[row] + dkh; iw = rowiw[row] + dkw;
+            if (p.reflect) {
+              ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
+            } else {
+              valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+            }
+          }
+          if (valid)
+            val = p.x[rowxb[row] + ((long)ih * p.W + iw) * p.Cin + ci];
+        }
+        As[row][kc] = val;
+      }
+    }
+
+    // ---- stage B^T (weight rows, k contiguous) ----
+    if (ALIGNED) {
+      #pragma unroll 2
+      for (int c = tid; c < BN * (BK / 8); c += NTHREADS) {
+        int n = c >> 3;
+        int kc = (c & 7) * 8;
+        long k = k0 + kc;
+        v8s val = {};
+        if (n0 + n < p.Cout && k < p.KTOT)
+          val = *(const v8s*)(p.w + (long)(n0 + n) * p.KTOT + k);
+        *(v8s*)&Bs[n][kc] = val;
+      }
+    } else {
+      for (int c = tid; c < BN * BK; c += NTHREADS) {
+        int n = c >> 6;
+        int kc = c & 63;
+        long k = k0 + kc;
+        Bs[n][kc] = (n0 + n < p.Cout && k < p.KTOT)
+                        ? p.w[(long)(n0 + n) * p.KTOT + k] : (short)0;
+      }
+    }
+    __syncthreads();
+
+    // ---- MFMA: 2 k-substeps of 32 ----
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 32) {
+      v8bf a0 = *(const v8bf*)&As[wm0 + 0 * 16 + fr][kk + fg * 8];
+      v8bf a1 = *(const v8bf*)&As[wm0 + 1 * 16 + fr][kk + fg * 8];
+      v8bf a2 = *(const v8bf*)&As[wm0 + 2 * 16 + fr][kk + fg * 8];
+      v8bf a3 = *(const v8bf*)&As[wm0 + 3 * 16 + fr][kk + fg * 8];
+      v8bf b0 = *(const v8bf*)&Bs[wn0 + 0 * 16 + fr][kk + fg * 8];
+      v8bf b1 = *(const v8bf*)&Bs[wn0 + 1 * 16 + fr][kk + fg * 8];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
+      acc[2][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b0, acc[2][0], 0, 0, 0);
+      acc[2][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b1, acc[2][1], 0, 0, 0);
+      acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b0, acc[3][0], 0, 0, 0);
+      acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b1, acc[3][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: bias + activation + bf16 store ----
+  #pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+    int n = n0 + wn0 + nf * 16 + fr;
+    if (n >= p.Cout) continue;
+    float bv = p.bias ? b2f(p.bias[n]) : 0.f;
+    #pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int rl = wm0 + mf * 16 + fg * 4 + r;
+        if (!rowok[rl]) continue;
+        float v = apply_act(acc[mf][nf][r] + bv, p.act, p.slope);
+        p.y[rowyb[rl] + n] = f2b(v);
+      }
+    }
+  }
+}
+
+// ---------------- weight gradient ----------------
+// dw[n][k] = Σ_m A[m][k] · dy[m][n];  A-tile and dy-tile staged TRANSPOSED
+// (m contiguous per row) so the MFMA reduce dim is m. fp32 atomics over
+// split-M slices.
+
+struct WgradParams {
+  const short* __restrict__ x;    // [B,H,W,Cin]
+  const short* __restrict__ dy;   // [B,OH,OW,Cout]
+  float* __restrict__ dw;         // [Cout,KH,KW,Cin] fp32 (pre-zeroed)
+  int B, H, W, Cin, OH, OW, Cout, KH, KW;
+  int stride, pt, pl, reflect;
+  long M, KTOT;
+  int ktiles, ntiles, slices;
+  long mchunks_per_slice;   // in units of 64 rows
+};
+
+constexpr int WG_BK = 64;   // k-tile (weight elements)
+constexpr int WG_BN = 64;   // n-tile (cout)
+constexpr int WG_BM = 64;   // m per iteration (the mfma reduce dim)
+constexpr int WG_LDM = WG_BM + 8;
+
+__global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
+  __shared__ short At[WG_BK][WG_LDM];   // [k][m]
+  __shared__ short Dt[WG_BN][WG_LDM];   // [n][m]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int kt = blockIdx.x % p.ktiles;
+  const int nt = (blockIdx.x / p.ktiles) % p.ntiles;
+  const int sl = blockIdx.x / (p.ktiles * p.ntiles);
+  const long k0 = (long)kt * WG_BK;
+  const int n0 = nt * WG_BN;
+  const long mstart = sl * p.mchunks_per_slice * WG_BM;
+  long mend = mstart + p.mchunks_per_slice * WG_BM;
+  if (mend > p.M) mend = p.M;
+
+  v4f acc[4][2] = {};
+  const int wr = wid >> 1, wc = wid & 1;
+  const int wk0 = wr * 64, wn0 = wc * 32;  // wave tile: 64 k x 32 n
+  const int fr = lane & 15, fg = lane >> 4;
+
+  // decode this thread's k lane once (for A staging): thread handles
+  // chunk c -> k-row = c>>3, m-chunk = (c&7)*8
+  for (long ms = mstart; ms < mend; ms += WG_BM) {
+    // ---- stage At[k][m] (scalar gather, transposing write) ----
+    for (int c = tid; c < WG_BK * 8; c += NTHREADS) {
+      int krow = c >> 3;
+      int mc = (c & 7) * 8;
+      long k = k0 + krow;
+      short vals[8];
+      int tap = 0, ci = 0, dkh = 0, dkw = 0;
+      bool krows_ok = k < p.KTOT;
+      if (krows_ok) {
+        tap = (int)(k / p.Cin);
+        ci = (int)(k - (long)tap * p.Cin);
+        dkh = tap / p.KW; dkw = tap - dkh * p.KW;
+      }
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        long m = ms + mc + j;
+        short v = 0;
+        if (krows_ok && m < p.M) {
+          int ow = (int)(m % p.OW);
+          int oh = (int)((m / p.OW) % p.OH);
+          int b = (int)(m / ((long)p.OW * p.OH));
+          int ih = oh * p.stride - p.pt + dkh;
+          int iw = ow * p.stride - p.pl + dkw;
+          bool valid = true;
+          if (p.reflect) {
+            ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
+          } else {
+            valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+          }
+          if (valid)
+            v = p.x[(((long)b * p.H + ih) * p.W + iw) * p.Cin + ci];
+        }
+        vals[j] = v;
+      }
+      *(v8s*)&At[krow][mc] = *(v8s*)vals;
+    }
+    // ---- stage Dt[n][m] ----
+    for (int c = tid; c < WG_BN * 8; c += NTHREADS) {
+      int n = c >> 3;
+      int mc = (c & 7) * 8;
+      short vals[8];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        long m = ms + mc + j;
+        vals[j] = (n0 + n < p.Cout && m < p.M)
+                      ? p.dy[m * p.Cout + n0 + n] : (short)0;
+      }
+      *(v8s*)&Dt[n][mc] = *(v8s*)vals;
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int kk = 0; kk < WG_BM; kk += 32) {
+      v8bf a0 = *(const v8bf*)&At[wk0 + 0 * 16 + fr][kk + fg * 8];
+      v8bf a1 = *(const v8bf*)&At[wk0 + 1 * 16 + fr][kk + fg * 8];
+      v8bf a2 = *(const v8bf*)&At[wk0 + 2 * 16 + fr][kk + fg * 8];
+      v8bf a3 = *(const v8bf*)&At[wk0 + 3 * 16 + fr][kk + fg * 8];
+      v8bf b0 = *(const v8bf*)&Dt[wn0 + 0 * 16 + fr][kk + fg * 8];
+      v8bf b1 = *(const v8bf*)&Dt[wn0 + 1 * 16 + fr][kk + fg * 8];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
+      acc[2][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b0, acc[2][0], 0, 0, 0);
+      acc[2][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b1, acc[2][1], 0, 0, 0);
+      acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b0, acc[3][0], 0, 0, 0);
+      acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b1, acc[3][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- atomic accumulate: D row = k (= (lane>>4)*4+r), col = n ----
+  #pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+    int n = n0 + wn0 + nf * 16 + fr;
+    if (n >= p.Cout) continue;
+    #pragma unroll
+    for (int kf = 0; kf < 4; ++kf) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long k = k0 + wk0 + kf * 16 + fg * 4 + r;
+        if (k < p.KTOT)
+          atomicAdd(&p.dw[(long)n * p.KTOT + k], acc[kf][nf][r]);
+      }
+    }
+  }
+}
+
+// ---------------- reflect fold (dgrad border scatter) ----------------
+// dx[b,i,j,c] = Σ over padded positions q with mirror(q - pad) == (i,j)
+__global__ void reflect_fold_kernel(const short* __restrict__ dxp,
+                                    float* __restrict__ dx_unused,
+                                    short* __restrict__ dx,
+                                    int B, int H, int W, int C,
+                                    int pt, int pb, int pl, int pr) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long total = (long)B * H * W * C;
+  if (idx >= total) return;
+  int c = (int)(idx % C);
+  long t = idx / C;
+  int j = (int)(t % W);
+  t /= W;
+  int i = (int)(t % H);
+  int b = (int)(t / H);
+  int HP = H + pt + pb, WP = W + pl + pr;
+  float s = 0.f;
+  // candidate padded rows: identity, top mirror, bottom mirror
+  int hc[3] = {pt + i, pt - i, pt + 2 * (H - 1) - i};
+  int wc_[3] = {pl + j, pl - j, pl + 2 * (W - 1) - j};
+  #pragma unroll
+  for (int a = 0; a < 3; ++a) {
+    int qh = hc[a];
+    if (qh < 0 || qh >= HP) continue;
+    if (a > 0 && qh == hc[0]) continue;           // dedupe (i==0 cases)
+    if (a == 2 && qh == hc[1]) continue;
+    if (mirror_idx(qh - pt, H) != i) continue;
+    #pragma unroll
+    for (int d = 0; d < 3; ++d) {
+      int qw = wc_[d];
+      if (qw < 0 || qw >= WP) continue;
+      if (d > 0 && qw == wc_[0]) continue;
+      if (d == 2 && qw == wc_[1]) continue;
+      if (mirror_idx(qw - pl, W) != j) continue;
+      s += b2f(dxp[(((long)b * HP + qh) * WP + qw) * C + c]);
+    }
+  }
+  dx[idx] = f2b(s);
+}
+
+// ================= host wrappers =================
+
+static inline int cdiv(long a, long b) { return (int)((a + b - 1) / b); }
+
+static void launch_conv(const ConvParams& p, bool is_convt, hipStream_t stream) {
+  dim3 grid(p.mtiles * p.ntiles);
+  bool aligned = (p.Cin % 8) == 0;
+  if (is_convt) {
+    if (aligned)
+      hipLaunchKernelGGL((conv_gemm_kernel<true, true>), grid, dim3(NTHREADS), 0, stream, p);
+    else
+      hipLaunchKernelGGL((conv_gemm_kernel<true, false>), grid, dim3(NTHREADS), 0, stream, p);
+  } else {
+    if (aligned)
+      hipLaunchKernelGGL((conv_gemm_kernel<false, true>), grid, dim3(NTHREADS), 0, stream, p);
+    else
+      hipLaunchKernelGGL((conv_gemm_kernel<false, false>), grid, dim3(NTHREADS), 0, stream, p);
+  }
+}
+
+static ConvParams fill_common(const at::Tensor& x, const at::Tensor& w,
+                              const c10::optional<at::Tensor>& bias,
+                              at::Tensor& y, int stride, int act, double slope) {
+  ConvParams p{};
+  p.x = (const short*)x.const_data_ptr();
+  p.w = (const short*)w.const_data_ptr();
+  p.bias = bias.has_value() ? (const short*)bias->const_data_ptr() : nullptr;
+  p.y = (short*)y.mutable_data_ptr();
+  p.B = x.size(0); p.H = x.size(1); p.W = x.size(2); p.Cin = x.size(3);
+  p.OH = y.size(1); p.OW = y.size(2); p.Cout = y.size(3);
+  p.KH = w.size(1); p.KW = w.size(2);
+  p.stride = stride;
+  p.act = act; p.slope = (float)slope;
+  p.M = (long)p.B * p.OH * p.OW;
+  p.KTOT = (long)p.KH * p.KW * p.Cin;
+  p.mtiles = cdiv(p.M, BM);
+  p.ntiles = cdiv(p.Cout, BN);
+  return p;
+}
+
+static void check_conv_inputs(const at::Tensor& x, const at::Tensor& w) {
+  TORCH_CHECK(x.is_cuda() && w.is_cuda(), "conv: tensors must be on GPU");
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 && w.scalar_type() == at::kBFloat16,
+              "conv: GPU path is bf16 (got ", x.scalar_type(), ")");
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous(), "conv: need contiguous");
+  TORCH_CHECK(w.size(3) == x.size(3), "conv: Cin mismatch");
+}
+
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias,
+                      int64_t stride, int64_t pt, int64_t pb, int64_t pl,
+                      int64_t pr, bool reflect, int64_t act, double slope) {
+  check_conv_inputs(x, w);
+  int H = x.size(1), W = x.size(2);
+  int KH = w.size(1), KW = w.size(2);
+  int OH = (H + pt + pb - KH) / stride + 1;
+  int OW = (W + pl + pr - KW) / stride + 1;
+  auto y = at::empty({x.size(0), OH, OW, w.size(0)}, x.options());
+  auto p = fill_common(x, w, bias, y, stride, act, slope);
+  p.pt = pt; p.pl = pl; p.reflect = reflect ? 1 : 0;
+  launch_conv(p, false, at::hip::getCurrentHIPStreamMasqueradingAsCUDA());
+  return y;
+}
+
+at::Tensor convt2d_fwd(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias,
+                       int64_t stride, int64_t pt, int64_t pl,
+                       int64_t out_h, int64_t out_w, int64_t act, double slope) {
+  check_conv_inputs(x, w);
+  auto y = at::empty({x.size(0), out_h, out_w, w.size(0)}, x.options());
+  auto p = fill_common(x, w, bias, y, stride, act, slope);
+  p.pt = pt; p.pl = pl; p.reflect = 0;
+  launch_conv(p, true, at::hip::getCurrentHIPStreamMasqueradingAsCUDA());
+  return y;
+}
+
+// conv dgrad: dx = gather-adjoint of conv_fwd == convt kernel on dy with the
+// channel-transposed weight (wt: [Cin,KH,KW,Cout]), same taps, no flip.
+at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt, int64_t H, int64_t W,
+                        int64_t stride, int64_t pt, int64_t pb, int64_t pl,
+                        int64_t pr, bool reflect) {
+  check_conv_inputs(dy, wt);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  if (!reflect) {
+    auto dx = at::empty({dy.size(0), H, W, wt.size(0)}, dy.options());
+    auto p = fill_common(dy, wt, c10::nullopt, dx, stride, ACT_NONE, 0.0);
+    p.pt = pt; p.pl = pl;
+    launch_conv(p, true, stream);
+    return dx;
+  }
+  // reflect: dgrad into the padded frame, then fold mirrors back
+  long HP = H + pt + pb, WP = W + pl + pr;
+  auto dxp = at::empty({dy.size(0), HP, WP, wt.size(0)}, dy.options());
+  auto p = fill_common(dy, wt, c10::nullopt, dxp, stride, ACT_NONE, 0.0);
+  p.pt = 0; p.pl = 0;
+  launch_conv(p, true, stream);
+  auto dx = at::empty({dy.size(0), H, W, wt.size(0)}, dy.options());
+  long total = dx.numel();
+  int threads = 256;
+  hipLaunchKernelGGL(reflect_fold_kernel, dim3(cdiv(total, threads)),
+                     dim3(threads), 0, stream,
+                     (const short*)dxp.const_data_ptr(), nullptr,
+                     (short*)dx.mutable_data_ptr(),
+                     (int)dy.size(0), (int)H, (int)W, (int)wt.size(0),
+                     (int)pt, (int)pb, (int)pl, (int)pr);
+  return dx;
+}
+
+// convt dgrad: adjoint of the convt gather == strided conv_fwd of dy.
+at::Tensor convt2d_dgrad(at::Tensor dy, at::Tensor wt, int64_t IH, int64_t IW,
+                         int64_t stride, int64_t pt, int64_t pl) {
+  check_conv_inputs(dy, wt);
+  int KH = wt.size(1), KW = wt.size(2);
+  auto dx = at::empty({dy.size(0), IH, IW, wt.size(0)}, dy.options());
+  auto p = fill_common(dy, wt, c10::nullopt, dx, stride, ACT_NONE, 0.0);
+  p.pt = pt; p.pl = pl; p.reflect = 0;
+  launch_conv(p, false, at::hip::getCurrentHIPStreamMasqueradingAsCUDA());
+  return dx;
+}
+
+at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
+                        int64_t stride, int64_t pt, int64_t pl, bool reflect) {
+  check_conv_inputs(x, dy.contiguous());
+  TORCH_CHECK(dy.size(0) == x.size(0));
+  WgradParams p{};
+  p.x = (const short*)x.const_data_ptr();
+  p.dy = (const short*)dy.const_data_ptr();
+  p.B = x.size(0); p.H = x.size(1); p.W = x.size(2); p.Cin = x.size(3);
+  p.OH = dy.size(1); p.OW = dy.size(2); p.Cout = dy.size(3);
+  p.KH = KH; p.KW = KW; p.stride = stride; p.pt = pt; p.pl = pl;
+  p.reflect = reflect ? 1 : 0;
+  p.M = (long)p.B * p.OH * p.OW;
+  p.KTOT = (long)KH * KW * p.Cin;
+  p.ktiles = cdiv(p.KTOT, WG_BK);
+  p.ntiles = cdiv(p.Cout, WG_BN);
+  // split M so total blocks ≈ 2-4x CU count
+  long mchunks = (p.M + WG_BM - 1) / WG_BM;
+  int target = std::max<long>(1, 512 / ((long)p.ktiles * p.ntiles));
+  int slices = (int)std::min<long>(mchunks, target);
+  p.mchunks_per_slice = (mchunks + slices - 1) / slices;
+  p.slices = slices;
+  auto dw = at::zeros({p.Cout, (long)KH, (long)KW, p.Cin},
+                      x.options().dtype(at::kFloat));
+  p.dw = (float*)dw.mutable_data_ptr();
+  dim3 grid((long)p.ktiles * p.ntiles * p.slices);
+  hipLaunchKernelGGL(wgrad_kernel, grid, dim3(NTHREADS), 0,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(), p);
+  return dw;
+}
+
+// ---- MFMA layout probe (test aid): C[16,16] = A[16,32] @ Bt[16,32]^T ----
+__global__ void mfma_probe_kernel(const short* a, const short* bt, float* c) {
+  int lane = threadIdx.x & 63;
+  int fr = lane & 15, fg = lane >> 4;
+  v8bf av = *(const v8bf*)(a + fr * 32 + fg * 8);
+  v8bf bv = *(const v8bf*)(bt + fr * 32 + fg * 8);
+  v4f d = {};
+  d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, d, 0, 0, 0);
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) c[(fg * 4 + r) * 16 + fr] = d[r];
+}
+
+at::Tensor mfma_probe(at::Tensor a, at::Tensor bt) {
+  TORCH_CHECK(a.sizes() == at::IntArrayRef({16, 32}) &&
+              bt.sizes() == at::IntArrayRef({16, 32}));
+  auto c = at::empty({16, 16}, a.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     (const short*)a.contiguous().const_data_ptr(),
+                     (const short*)bt.contiguous().const_data_ptr(),
+                     (float*)c.mutable_data_ptr());
+  return c;
+}
+
+}  // namespace cyg

@@ -33,7 +33,7 @@ import torch
 import torch.nn.functional as F
 
 from . import backend
-from .shadow import compute_weight
+from .shadow import compute_weight, compute_weight_t
 
 ACT_NONE, ACT_RELU, ACT_LRELU, ACT_TANH = 0, 1, 2, 3
 
@@ -110,16 +110,19 @@ class _ConvFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, w, y = ctx.saved_tensors
         stride, pads, reflect, act, slope, has_bias = ctx.conf
+        pt, pb, pl, pr = pads
         ext = backend.ext()
         dy = dy.contiguous()
         if act != ACT_NONE:
             dy = ext.act_bwd(dy, y, act, slope)
-        wc = compute_weight(w, x)
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            dx = ext.conv2d_dgrad(dy, wc, x.shape[1], x.shape[2], stride, *pads, reflect)
+            wt = compute_weight_t(w, x)
+            dx = ext.conv2d_dgrad(dy, wt, x.shape[1], x.shape[2], stride,
+                                  pt, pb, pl, pr, reflect)
         if ctx.needs_input_grad[1]:
-            dw = ext.conv2d_wgrad(x, dy, w.shape[1], w.shape[2], stride, *pads, reflect)
+            dw = ext.conv2d_wgrad(x, dy, w.shape[1], w.shape[2], stride,
+                                  pt, pl, reflect)
             if dw.dtype != w.dtype:
                 dw = dw.to(w.dtype)
         if has_bias and ctx.needs_input_grad[2]:
@@ -146,13 +149,17 @@ class _ConvTFn(torch.autograd.Function):
         dy = dy.contiguous()
         if act != ACT_NONE:
             dy = ext.act_bwd(dy, y, act, slope)
-        wc = compute_weight(w, x)
         dx = dw = db = None
-        # adjoint of the gather is the forward strided conv
         if ctx.needs_input_grad[0]:
-            dx = ext.convt2d_dgrad(dy, wc, x.shape[1], x.shape[2], stride, pt, pl)
+            # adjoint of the gather is the forward strided conv of dy
+            wt = compute_weight_t(w, x)
+            dx = ext.convt2d_dgrad(dy, wt, x.shape[1], x.shape[2], stride, pt, pl)
         if ctx.needs_input_grad[1]:
-            dw = ext.convt2d_wgrad(x, dy, w.shape[1], w.shape[2], stride, pt, pl)
+            # convT wgrad == conv wgrad with roles swapped (x := dy, dy := x),
+            # then a channel transpose back to OHWI
+            dwc = ext.conv2d_wgrad(dy, x, w.shape[1], w.shape[2], stride,
+                                   pt, pl, False)
+            dw = dwc.permute(3, 1, 2, 0).contiguous()
             if dw.dtype != w.dtype:
                 dw = dw.to(w.dtype)
         if has_bias and ctx.needs_input_grad[2]:

@@ -31,8 +31,9 @@ class _PerSampleLossFn(torch.autograd.Function):
         y_true, y_pred = ctx.saved_tensors
         gt, gp = backend.ext().persample_loss_bwd(
             y_true, y_pred, dout.contiguous().float(), ctx.squared,
-            ctx.needs_input_grad[0], ctx.needs_input_grad[1])
-        return gt, gp, None
+            True, True)
+        return (gt if ctx.needs_input_grad[0] else None,
+                gp if ctx.needs_input_grad[1] else None, None)
 
 
 def _per_sample_ref(y_true, y_pred, squared: bool) -> torch.Tensor:

@@ -14,6 +14,7 @@ import torch
 from torch.utils.weak import WeakTensorKeyDictionary
 
 _cache: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
+_cache_t: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
 
 
 def bf16_shadow(t: torch.Tensor) -> torch.Tensor:
@@ -36,3 +37,21 @@ def compute_weight(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
     if w.dtype != like.dtype:
         return w.detach().to(like.dtype)
     return w.detach()
+
+
+def compute_weight_t(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
+    """Channel-transposed weight (OHWI [O,kh,kw,I] -> [I,kh,kw,O]) in the
+    compute dtype — the B^T operand of the adjoint (dgrad) kernels. Cached
+    per master-version like the bf16 shadow (the backward of each of the
+    generator's 3 calls per step reuses it)."""
+    ent = _cache_t.get(w)
+    ver = w._version
+    if ent is not None and ent[0] == ver and ent[1] == like.dtype:
+        return ent[2]
+    wt = w.detach().permute(3, 1, 2, 0).contiguous()
+    if like.dtype == torch.bfloat16 and wt.dtype != torch.bfloat16:
+        wt = wt.to(torch.bfloat16)
+    elif wt.dtype != like.dtype:
+        wt = wt.to(like.dtype)
+    _cache_t[w] = (ver, like.dtype, wt)
+    return wt

@@ -1,0 +1,79 @@
+"""End-to-end GPU training step through the HIP kernel path."""
+
+import argparse
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def make_args(tmp_path, batch=2, nrb=2):
+    a = argparse.Namespace()
+    a.output_dir = str(tmp_path)
+    a.batch_size = batch
+    a.global_batch_size = batch
+    a.num_residual_blocks = nrb
+    a.compute_dtype = torch.bfloat16
+    return a
+
+
+def test_train_step_bf16(tmp_path):
+    from cyclegan_amd.parallel import DistContext
+    from cyclegan_amd.trainer import CycleGAN
+    torch.manual_seed(0)
+    ctx = DistContext(device=torch.device("cuda", 0))
+    gan = CycleGAN(make_args(tmp_path), ctx)
+    x = torch.rand(2, 64, 64, 3, device=ctx.device, dtype=torch.bfloat16) * 2 - 1
+    y = torch.rand(2, 64, 64, 3, device=ctx.device, dtype=torch.bfloat16) * 2 - 1
+    r0 = gan.train_step(x, y)
+    torch.cuda.synchronize()
+    for k, v in r0.items():
+        assert torch.isfinite(v), k
+    for _ in range(12):
+        r = gan.train_step(x, y)
+    torch.cuda.synchronize()
+    assert r["loss_G/cycle"].item() < r0["loss_G/cycle"].item()
+    assert r["loss_F/cycle"].item() < r0["loss_F/cycle"].item()
+    for g in gan.groups.values():
+        assert g.check_views()
+
+
+def test_gpu_step_matches_cpu_fp32_closely(tmp_path):
+    """One step, same weights+data: bf16 HIP losses must track the fp32 CPU
+    reference losses to bf16 tolerance (catches systematic kernel bias)."""
+    from cyclegan_amd.parallel import DistContext
+    from cyclegan_amd.trainer import CycleGAN
+    g = torch.Generator().manual_seed(5)
+    x = torch.rand(1, 64, 64, 3, generator=g) * 2 - 1
+    y = torch.rand(1, 64, 64, 3, generator=g) * 2 - 1
+
+    torch.manual_seed(7)
+    ctx_g = DistContext(device=torch.device("cuda", 0))
+    gan_g = CycleGAN(make_args(tmp_path, batch=1), ctx_g)
+    r_gpu = gan_g.test_step(x, y)
+
+    torch.manual_seed(7)
+    ctx_c = DistContext(device=torch.device("cpu"))
+    args = make_args(tmp_path, batch=1)
+    args.compute_dtype = torch.float32
+    gan_c = CycleGAN(args, ctx_c)
+    r_cpu = gan_c.test_step(x, y)
+
+    for k in r_cpu:
+        a, b = r_gpu[k].item(), r_cpu[k].item()
+        assert abs(a - b) <= 0.05 * (abs(b) + 0.05), (k, a, b)
+
+
+def test_checkpoint_roundtrip_gpu(tmp_path):
+    from cyclegan_amd.parallel import DistContext
+    from cyclegan_amd.trainer import CycleGAN
+    torch.manual_seed(0)
+    ctx = DistContext(device=torch.device("cuda", 0))
+    gan = CycleGAN(make_args(tmp_path), ctx)
+    x = torch.rand(1, 64, 64, 3, device=ctx.device, dtype=torch.bfloat16)
+    gan.train_step(x, x)
+    gan.save_checkpoint()
+    gan2 = CycleGAN(make_args(tmp_path), ctx)
+    assert gan2.load_checkpoint()
+    assert torch.equal(gan.groups["G"].flat_param, gan2.groups["G"].flat_param)

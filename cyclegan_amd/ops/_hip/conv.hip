@@ -236,7 +236,7 @@ struct WgradParams {
   long mchunks_per_slice;   // in units of 64 rows
 };
 
-constexpr int WG_BK = 64;   // k-tile (weight elements)
+constexpr int WG_BK = 128;  // k-tile (weight elements; 2x2 waves x 64k each)
 constexpr int WG_BN = 64;   // n-tile (cout)
 constexpr int WG_BM = 64;   // m per iteration (the mfma reduce dim)
 constexpr int WG_LDM = WG_BM + 8;

@@ -237,7 +237,7 @@ struct WgradParams {
   long mchunks_per_slice;   // in units of 64 rows
 };
 
-constexpr int WG_BK = 64;   // k-tile (weight elements)
+constexpr int WG_BK = 128;  // k-tile (weight elements; 2x2 waves x 64k each)
 constexpr int WG_BN = 64;   // n-tile (cout)
 constexpr int WG_BM = 64;   // m per iteration (the mfma reduce dim)
 constexpr int WG_LDM = WG_BM + 8;
@@ -515,7 +515,8 @@ at::Tensor convt2d_dgrad(at::Tensor dy, at::Tensor wt, int64_t IH, int64_t IW,
 
 at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
                         int64_t stride, int64_t pt, int64_t pl, bool reflect) {
-  check_conv_inputs(x, dy.contiguous());
+  TORCH_CHECK(x.is_cuda() && dy.is_cuda() && x.is_contiguous() && dy.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 && dy.scalar_type() == at::kBFloat16);
   TORCH_CHECK(dy.size(0) == x.size(0));
   WgradParams p{};
   p.x = (const short*)x.const_data_ptr();

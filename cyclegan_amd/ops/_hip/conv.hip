@@ -262,27 +262,26 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
   const int wk0 = wr * 64, wn0 = wc * 32;  // wave tile: 64 k x 32 n
   const int fr = lane & 15, fg = lane >> 4;
 
-  // decode this thread's k lane once (for A staging): thread handles
-  // chunk c -> k-row = c>>3, m-chunk = (c&7)*8
+  // Staging is transposed: global loads VECTORIZE along the natural inner
+  // dims (ci for x, cout for dy) and scatter 8 scalar ds_writes into the
+  // [k][m] / [n][m] images (LDS scatter ≪ scalar global gather).
+  const bool a_vec = (p.Cin % 8) == 0;
+  const bool d_vec = (p.Cout % 8) == 0;
+
   for (long ms = mstart; ms < mend; ms += WG_BM) {
-    // ---- stage At[k][m] (scalar gather, transposing write) ----
-    for (int c = tid; c < WG_BK * 8; c += NTHREADS) {
-      int krow = c >> 3;
-      int mc = (c & 7) * 8;
-      long k = k0 + krow;
-      short vals[8];
-      int tap = 0, ci = 0, dkh = 0, dkw = 0;
-      bool krows_ok = k < p.KTOT;
-      if (krows_ok) {
-        tap = (int)(k / p.Cin);
-        ci = (int)(k - (long)tap * p.Cin);
-        dkh = tap / p.KW; dkw = tap - dkh * p.KW;
-      }
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        long m = ms + mc + j;
-        short v = 0;
-        if (krows_ok && m < p.M) {
+    __syncthreads();  // previous iteration's MFMA reads done
+    // ---- stage At[k][m]: chunk = (m, 8 k-rows) ----
+    if (a_vec) {
+      for (int c = tid; c < WG_BM * (WG_BK / 8); c += NTHREADS) {
+        int m_loc = c % WG_BM;
+        int kc = (c / WG_BM) * 8;      // first of 8 consecutive k (same tap)
+        long k = k0 + kc;
+        v8s val = {};
+        long m = ms + m_loc;
+        if (k < p.KTOT && m < p.M) {
+          int tap = (int)(k / p.Cin);
+          int ci = (int)(k - (long)tap * p.Cin);
+          int dkh = tap / p.KW, dkw = tap - (tap / p.KW) * p.KW;
           int ow = (int)(m % p.OW);
           int oh = (int)((m / p.OW) % p.OH);
           int b = (int)(m / ((long)p.OW * p.OH));
@@ -295,24 +294,73 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
             valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
           }
           if (valid)
-            v = p.x[(((long)b * p.H + ih) * p.W + iw) * p.Cin + ci];
+            val = *(const v8s*)(p.x + (((long)b * p.H + ih) * p.W + iw) * p.Cin + ci);
         }
-        vals[j] = v;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) At[kc + j][m_loc] = val[j];
       }
-      *(v8s*)&At[krow][mc] = *(v8s*)vals;
+    } else {
+      for (int c = tid; c < WG_BK * 8; c += NTHREADS) {
+        int krow = c >> 3;
+        int mc = (c & 7) * 8;
+        long k = k0 + krow;
+        short vals[8];
+        int tap = 0, ci = 0, dkh = 0, dkw = 0;
+        bool krows_ok = k < p.KTOT;
+        if (krows_ok) {
+          tap = (int)(k / p.Cin);
+          ci = (int)(k - (long)tap * p.Cin);
+          dkh = tap / p.KW; dkw = tap - dkh * p.KW;
+        }
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          long m = ms + mc + j;
+          short v = 0;
+          if (krows_ok && m < p.M) {
+            int ow = (int)(m % p.OW);
+            int oh = (int)((m / p.OW) % p.OH);
+            int b = (int)(m / ((long)p.OW * p.OH));
+            int ih = oh * p.stride - p.pt + dkh;
+            int iw = ow * p.stride - p.pl + dkw;
+            bool valid = true;
+            if (p.reflect) {
+              ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
+            } else {
+              valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+            }
+            if (valid)
+              v = p.x[(((long)b * p.H + ih) * p.W + iw) * p.Cin + ci];
+          }
+          vals[j] = v;
+        }
+        *(v8s*)&At[krow][mc] = *(v8s*)vals;
+      }
     }
-    // ---- stage Dt[n][m] ----
-    for (int c = tid; c < WG_BN * 8; c += NTHREADS) {
-      int n = c >> 3;
-      int mc = (c & 7) * 8;
-      short vals[8];
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        long m = ms + mc + j;
-        vals[j] = (n0 + n < p.Cout && m < p.M)
-                      ? p.dy[m * p.Cout + n0 + n] : (short)0;
+    // ---- stage Dt[n][m]: chunk = (m, 8 n) ----
+    if (d_vec) {
+      for (int c = tid; c < WG_BM * (WG_BN / 8); c += NTHREADS) {
+        int m_loc = c % WG_BM;
+        int nc = (c / WG_BM) * 8;
+        long m = ms + m_loc;
+        v8s val = {};
+        if (m < p.M && n0 + nc < p.Cout)
+          val = *(const v8s*)(p.dy + m * p.Cout + n0 + nc);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) Dt[nc + j][m_loc] = val[j];
       }
-      *(v8s*)&Dt[n][mc] = *(v8s*)vals;
+    } else {
+      for (int c = tid; c < WG_BN * 8; c += NTHREADS) {
+        int n = c >> 3;
+        int mc = (c & 7) * 8;
+        short vals[8];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          long m = ms + mc + j;
+          vals[j] = (n0 + n < p.Cout && m < p.M)
+                        ? p.dy[m * p.Cout + n0 + n] : (short)0;
+        }
+        *(v8s*)&Dt[n][mc] = *(v8s*)vals;
+      }
     }
     __syncthreads();
 

@@ -18,11 +18,11 @@ constexpr int NT = 256;
 
 static inline int cdiv64(long a, long b) { return (int)((a + b - 1) / b); }
 
-// ---- pass 1: partial sum / sumsq over a spatial slice ----
-// grid (B * S), each block covers rows [slice] of HW; C % 8 == 0.
+// ---- pass 1: slab partials over spatial slices (no atomics) ----
+// grid (B * S); writes psum/psq[(sl*B + b)*C + c]; C % 8 == 0, C/8 <= NT.
 __global__ __launch_bounds__(NT) void in_reduce_kernel(
-    const short* __restrict__ x, float* __restrict__ sum,
-    float* __restrict__ sumsq, int B, long HW, int C, int S) {
+    const short* __restrict__ x, float* __restrict__ psum,
+    float* __restrict__ psq, int B, long HW, int C, int S) {
   int b = blockIdx.x / S;
   int sl = blockIdx.x % S;
   long rows = (HW + S - 1) / S;
@@ -31,69 +31,54 @@ __global__ __launch_bounds__(NT) void in_reduce_kernel(
   const int tid = threadIdx.x;
   const short* xb = x + (long)b * HW * C;
 
-  // thread handles channel-group g = tid % gpr, rows strided by NT/gpr
-  if (gpr <= NT) {
-    int g = tid % gpr;
-    int rstep = NT / gpr;
-    int rof = tid / gpr;
-    float s[8] = {}, q[8] = {};
-    for (long r = r0 + rof; r < r1; r += rstep) {
-      v8s v = *(const v8s*)(xb + r * C + g * 8);
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float f = b2f(v[j]);
-        s[j] += f; q[j] += f * f;
-      }
-    }
-    // reduce across the rstep threads sharing this channel group via LDS
-    __shared__ float red[NT * 2];
+  int g = tid % gpr;
+  int rstep = NT / gpr;
+  int rof = tid / gpr;
+  float s[8] = {}, q[8] = {};
+  for (long r = r0 + rof; r < r1; r += rstep) {
+    v8s v = *(const v8s*)(xb + r * C + g * 8);
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      red[tid] = s[j];
-      red[NT + tid] = q[j];
-      __syncthreads();
-      if (rof == 0) {
-        float ts = 0, tq = 0;
-        for (int k = 0; k < rstep; ++k) {
-          ts += red[g + k * gpr];
-          tq += red[NT + g + k * gpr];
-        }
-        atomicAdd(&sum[(long)b * C + g * 8 + j], ts);
-        atomicAdd(&sumsq[(long)b * C + g * 8 + j], tq);
-      }
-      __syncthreads();
+      float f = b2f(v[j]);
+      s[j] += f; q[j] += f * f;
     }
-  } else {
-    // C > 2048: each thread walks channel groups
-    for (int g = tid; g < gpr; g += NT) {
-      float s[8] = {}, q[8] = {};
-      for (long r = r0; r < r1; ++r) {
-        v8s v = *(const v8s*)(xb + r * C + g * 8);
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float f = b2f(v[j]);
-          s[j] += f; q[j] += f * f;
-        }
+  }
+  // reduce across the rstep threads sharing this channel group via LDS
+  __shared__ float red[NT * 2];
+  long out = ((long)sl * B + b) * C + g * 8;
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[tid] = s[j];
+    red[NT + tid] = q[j];
+    __syncthreads();
+    if (rof == 0) {
+      float ts = 0, tq = 0;
+      for (int k = 0; k < rstep; ++k) {
+        ts += red[g + k * gpr];
+        tq += red[NT + g + k * gpr];
       }
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        atomicAdd(&sum[(long)b * C + g * 8 + j], s[j]);
-        atomicAdd(&sumsq[(long)b * C + g * 8 + j], q[j]);
-      }
+      psum[out + j] = ts;
+      psq[out + j] = tq;
     }
+    __syncthreads();
   }
 }
 
-// ---- pass 2: finalize mean / rstd ----
-__global__ void in_finalize_kernel(float* __restrict__ sum,
-                                   float* __restrict__ sumsq,
+// ---- pass 2: finalize mean / rstd from the S slabs ----
+__global__ void in_finalize_kernel(const float* __restrict__ psum,
+                                   const float* __restrict__ psq,
                                    float* __restrict__ mean,
                                    float* __restrict__ rstd,
-                                   long n, float inv_hw, float eps) {
+                                   long n, int S, float inv_hw, float eps) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
-  float m = sum[i] * inv_hw;
-  float var = sumsq[i] * inv_hw - m * m;
+  float s = 0, q = 0;
+  for (int k = 0; k < S; ++k) {
+    s += psum[(long)k * n + i];
+    q += psq[(long)k * n + i];
+  }
+  float m = s * inv_hw;
+  float var = q * inv_hw - m * m;
   if (var < 0.f) var = 0.f;
   mean[i] = m;
   rstd[i] = rsqrtf(var + eps);
@@ -135,11 +120,11 @@ __global__ __launch_bounds__(NT) void in_norm_kernel(
   }
 }
 
-// ---- backward pass 1: s1 = Σ dy, s2 = Σ dy*xhat per (b,c) ----
+// ---- backward pass 1: slab partials of s1 = Σ dy, s2 = Σ dy*xhat ----
 __global__ __launch_bounds__(NT) void in_bwd_reduce_kernel(
     const short* __restrict__ dy, const short* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    float* __restrict__ s1, float* __restrict__ s2, int B, long HW, int C,
+    float* __restrict__ p1, float* __restrict__ p2, int B, long HW, int C,
     int S) {
   int b = blockIdx.x / S;
   int sl = blockIdx.x % S;
@@ -149,11 +134,9 @@ __global__ __launch_bounds__(NT) void in_bwd_reduce_kernel(
   const int tid = threadIdx.x;
   const long base = (long)b * HW * C;
 
-  // thread accumulates locally for its fixed channel-group, atomics once
   int g = tid % gpr;
-  int rstep = max(1, NT / gpr);
+  int rstep = NT / gpr;
   int rof = tid / gpr;
-  if (rof >= rstep) return;
   float a1[8] = {}, a2[8] = {};
   float mv[8], rv[8];
   #pragma unroll
@@ -172,11 +155,41 @@ __global__ __launch_bounds__(NT) void in_bwd_reduce_kernel(
       a1[j] += d; a2[j] += d * xh;
     }
   }
+  __shared__ float red[NT * 2];
+  long out = ((long)sl * B + b) * C + g * 8;
   #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    atomicAdd(&s1[(long)b * C + g * 8 + j], a1[j]);
-    atomicAdd(&s2[(long)b * C + g * 8 + j], a2[j]);
+    red[tid] = a1[j];
+    red[NT + tid] = a2[j];
+    __syncthreads();
+    if (rof == 0) {
+      float t1 = 0, t2 = 0;
+      for (int k = 0; k < rstep; ++k) {
+        t1 += red[g + k * gpr];
+        t2 += red[NT + g + k * gpr];
+      }
+      p1[out + j] = t1;
+      p2[out + j] = t2;
+    }
+    __syncthreads();
   }
+}
+
+// ---- backward: sum the S slabs into s1/s2 [B,C] ----
+__global__ void in_bwd_finalize_kernel(const float* __restrict__ p1,
+                                       const float* __restrict__ p2,
+                                       float* __restrict__ s1,
+                                       float* __restrict__ s2, long n,
+                                       int S) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  float t1 = 0, t2 = 0;
+  for (int k = 0; k < S; ++k) {
+    t1 += p1[(long)k * n + i];
+    t2 += p2[(long)k * n + i];
+  }
+  s1[i] = t1;
+  s2[i] = t2;
 }
 
 // ---- backward pass 2: dx; also dgamma/dbeta reduce over b ----
@@ -419,22 +432,22 @@ std::vector<at::Tensor> instnorm_fwd(at::Tensor x, at::Tensor gamma,
               "instnorm: unsupported channel count ", C);
   auto stream = at::cuda::getCurrentCUDAStream();
   auto fopt = x.options().dtype(at::kFloat);
-  auto sum = at::zeros({B, C}, fopt);
-  auto sumsq = at::zeros({B, C}, fopt);
+  int S = slices_for(HW, B);
+  auto psum = at::empty({S, B, C}, fopt);
+  auto psq = at::empty({S, B, C}, fopt);
   auto mean = at::empty({B, C}, fopt);
   auto rstd = at::empty({B, C}, fopt);
   auto y = at::empty_like(x);
-  int S = slices_for(HW, B);
   hipLaunchKernelGGL(in_reduce_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)x.const_data_ptr(),
-                     (float*)sum.mutable_data_ptr(),
-                     (float*)sumsq.mutable_data_ptr(), B, HW, C, S);
+                     (float*)psum.mutable_data_ptr(),
+                     (float*)psq.mutable_data_ptr(), B, HW, C, S);
   long n = (long)B * C;
   hipLaunchKernelGGL(in_finalize_kernel, dim3(cdiv64(n, 256)), dim3(256), 0,
-                     stream, (float*)sum.mutable_data_ptr(),
-                     (float*)sumsq.mutable_data_ptr(),
+                     stream, (const float*)psum.const_data_ptr(),
+                     (const float*)psq.const_data_ptr(),
                      (float*)mean.mutable_data_ptr(),
-                     (float*)rstd.mutable_data_ptr(), n, 1.f / (float)HW,
+                     (float*)rstd.mutable_data_ptr(), n, S, 1.f / (float)HW,
                      (float)eps);
   const short* res = residual.has_value()
                          ? (const short*)residual->const_data_ptr() : nullptr;
@@ -456,21 +469,32 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
               dy.is_contiguous() && x.is_contiguous());
   int B = x.size(0), C = x.size(3);
   long HW = (long)x.size(1) * x.size(2);
+  TORCH_CHECK(C % 8 == 0 && (C / 8) <= 256 && 256 % std::min(C / 8, 256) == 0,
+              "instnorm_bwd: unsupported channel count ", C);
   auto stream = at::cuda::getCurrentCUDAStream();
   auto fopt = x.options().dtype(at::kFloat);
-  auto s1 = at::zeros({B, C}, fopt);
-  auto s2 = at::zeros({B, C}, fopt);
+  int S = slices_for(HW, B);
+  auto p1 = at::empty({S, B, C}, fopt);
+  auto p2 = at::empty({S, B, C}, fopt);
+  auto s1 = at::empty({B, C}, fopt);
+  auto s2 = at::empty({B, C}, fopt);
   auto dx = at::empty_like(x);
   auto dgamma = at::empty({C}, fopt);
   auto dbeta = at::empty({C}, fopt);
-  int S = slices_for(HW, B);
   hipLaunchKernelGGL(in_bwd_reduce_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)dy.const_data_ptr(),
                      (const short*)x.const_data_ptr(),
                      (const float*)mean.const_data_ptr(),
                      (const float*)rstd.const_data_ptr(),
+                     (float*)p1.mutable_data_ptr(),
+                     (float*)p2.mutable_data_ptr(), B, HW, C, S);
+  long nbc = (long)B * C;
+  hipLaunchKernelGGL(in_bwd_finalize_kernel, dim3(cdiv64(nbc, 256)),
+                     dim3(256), 0, stream,
+                     (const float*)p1.const_data_ptr(),
+                     (const float*)p2.const_data_ptr(),
                      (float*)s1.mutable_data_ptr(),
-                     (float*)s2.mutable_data_ptr(), B, HW, C, S);
+                     (float*)s2.mutable_data_ptr(), nbc, S);
   hipLaunchKernelGGL(in_bwd_dx_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)dy.const_data_ptr(),
                      (const short*)x.const_data_ptr(),

@@ -46,8 +46,9 @@ struct ConvParams {
 // ---------------- shared GEMM core ----------------
 // As rows = m (output pixel), Bs rows = n (cout); both k-contiguous.
 
-template <bool IS_CONVT, bool ALIGNED>
+template <bool IS_CONVT, bool ALIGNED, int STRIDE>
 __global__ __launch_bounds__(NTHREADS) void conv_gemm_kernel(ConvParams p) {
+  const int stride = STRIDE ? STRIDE : p.stride;
   __shared__ short As[BM][LDK];
   __shared__ short Bs[BN][LDK];
   __shared__ long rowxb[BM];    // batch base offset into x
@@ -78,8 +79,8 @@ __global__ __launch_bounds__(NTHREADS) void conv_gemm_kernel(ConvParams p) {
       rowih[r] = oh + p.pt;   // output coord + pad (gather: o = (i+pt-dk)/s)
       rowiw[r] = ow + p.pl;
     } else {
-      rowih[r] = oh * p.stride - p.pt;
-      rowiw[r] = ow * p.stride - p.pl;
+      rowih[r] = oh * stride - p.pt;
+      rowiw[r] = ow * stride - p.pl;
     }
   }
   __syncthreads();
@@ -108,9 +109,9 @@ __global__ __launch_bounds__(NTHREADS) void conv_gemm_kernel(ConvParams p) {
           int ih, iw;
           if (IS_CONVT) {
             int nh = rowih[row] - dkh, nw = rowiw[row] - dkw;
-            valid = nh >= 0 && nw >= 0 && (nh % p.stride) == 0 &&
-                    (nw % p.stride) == 0;
-            ih = nh / p.stride; iw = nw / p.stride;
+            valid = nh >= 0 && nw >= 0 && (nh % stride) == 0 &&
+                    (nw % stride) == 0;
+            ih = nh / stride; iw = nw / stride;
             valid = valid && ih < p.H && iw < p.W;
           } else {
             ih = rowih[row] + dkh; iw = rowiw[row] + dkw;
@@ -139,9 +140,9 @@ __global__ __launch_bounds__(NTHREADS) void conv_gemm_kernel(ConvParams p) {
           int ih, iw;
           if (IS_CONVT) {
             int nh = rowih[row] - dkh, nw = rowiw[row] - dkw;
-            valid = nh >= 0 && nw >= 0 && (nh % p.stride) == 0 &&
-                    (nw % p.stride) == 0;
-            ih = nh / p.stride; iw = nw / p.stride;
+            valid = nh >= 0 && nw >= 0 && (nh % stride) == 0 &&
+                    (nw % stride) == 0;
+            ih = nh / stride; iw = nw / stride;
             valid = valid && ih < p.H && iw < p.W;
           } else {
             ih = rowih[row] + dkh; iw = rowiw[row] + dkw;
@@ -447,19 +448,32 @@ __global__ void reflect_fold_kernel(const short* __restrict__ dxp,
 
 static inline int cdiv(long a, long b) { return (int)((a + b - 1) / b); }
 
+template <bool IS_CONVT, bool ALIGNED>
+static void launch_conv_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
+  switch (p.stride) {
+    case 1:
+      hipLaunchKernelGGL((conv_gemm_kernel<IS_CONVT, ALIGNED, 1>), grid,
+                         dim3(NTHREADS), 0, stream, p);
+      break;
+    case 2:
+      hipLaunchKernelGGL((conv_gemm_kernel<IS_CONVT, ALIGNED, 2>), grid,
+                         dim3(NTHREADS), 0, stream, p);
+      break;
+    default:
+      hipLaunchKernelGGL((conv_gemm_kernel<IS_CONVT, ALIGNED, 0>), grid,
+                         dim3(NTHREADS), 0, stream, p);
+  }
+}
+
 static void launch_conv(const ConvParams& p, bool is_convt, hipStream_t stream) {
   dim3 grid(p.mtiles * p.ntiles);
   bool aligned = (p.Cin % 8) == 0;
   if (is_convt) {
-    if (aligned)
-      hipLaunchKernelGGL((conv_gemm_kernel<true, true>), grid, dim3(NTHREADS), 0, stream, p);
-    else
-      hipLaunchKernelGGL((conv_gemm_kernel<true, false>), grid, dim3(NTHREADS), 0, stream, p);
+    if (aligned) launch_conv_s<true, true>(p, grid, stream);
+    else launch_conv_s<true, false>(p, grid, stream);
   } else {
-    if (aligned)
-      hipLaunchKernelGGL((conv_gemm_kernel<false, true>), grid, dim3(NTHREADS), 0, stream, p);
-    else
-      hipLaunchKernelGGL((conv_gemm_kernel<false, false>), grid, dim3(NTHREADS), 0, stream, p);
+    if (aligned) launch_conv_s<false, true>(p, grid, stream);
+    else launch_conv_s<false, false>(p, grid, stream);
   }
 }
 

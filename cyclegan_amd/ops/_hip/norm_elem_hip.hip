@@ -65,33 +65,18 @@ __global__ __launch_bounds__(NT) void in_reduce_kernel(
   }
 }
 
-// ---- pass 2: finalize mean / rstd from the S slabs ----
-__global__ void in_finalize_kernel(const float* __restrict__ psum,
-                                   const float* __restrict__ psq,
-                                   float* __restrict__ mean,
-                                   float* __restrict__ rstd,
-                                   long n, int S, float inv_hw, float eps) {
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= n) return;
-  float s = 0, q = 0;
-  for (int k = 0; k < S; ++k) {
-    s += psum[(long)k * n + i];
-    q += psq[(long)k * n + i];
-  }
-  float m = s * inv_hw;
-  float var = q * inv_hw - m * m;
-  if (var < 0.f) var = 0.f;
-  mean[i] = m;
-  rstd[i] = rsqrtf(var + eps);
-}
 
-// ---- pass 3: normalize + affine + act (+ residual) ----
+
+// ---- pass 2 (fused): per-block slab-sum stats + normalize + affine +
+// act (+ residual); slice-0 blocks also persist mean/rstd for backward ----
+constexpr int MAXC = 2048;
 __global__ __launch_bounds__(NT) void in_norm_kernel(
     const short* __restrict__ x, const float* __restrict__ gamma,
-    const float* __restrict__ beta, const float* __restrict__ mean,
-    const float* __restrict__ rstd, const short* __restrict__ res,
+    const float* __restrict__ beta, const float* __restrict__ psum,
+    const float* __restrict__ psq, float* __restrict__ mean,
+    float* __restrict__ rstd, const short* __restrict__ res,
     short* __restrict__ y, int B, long HW, int C, int S, int act,
-    float slope) {
+    float slope, float eps) {
   int b = blockIdx.x / S;
   int sl = blockIdx.x % S;
   long rows = (HW + S - 1) / S;
@@ -99,6 +84,27 @@ __global__ __launch_bounds__(NT) void in_norm_kernel(
   const int gpr = C / 8;
   const int tid = threadIdx.x;
   const long base = (long)b * HW * C;
+  const float inv_hw = 1.f / (float)HW;
+
+  __shared__ float sm[MAXC], sr[MAXC];
+  for (int c = tid; c < C; c += NT) {
+    float sv = 0, qv = 0;
+    for (int k = 0; k < S; ++k) {
+      sv += psum[((long)k * B + b) * C + c];
+      qv += psq[((long)k * B + b) * C + c];
+    }
+    float m = sv * inv_hw;
+    float var = qv * inv_hw - m * m;
+    if (var < 0.f) var = 0.f;
+    float rs = rsqrtf(var + eps);
+    sm[c] = m;
+    sr[c] = rs;
+    if (sl == 0) {
+      mean[(long)b * C + c] = m;
+      rstd[(long)b * C + c] = rs;
+    }
+  }
+  __syncthreads();
 
   for (long e = tid; e < (r1 - r0) * gpr; e += NT) {
     long r = r0 + e / gpr;
@@ -111,9 +117,7 @@ __global__ __launch_bounds__(NT) void in_norm_kernel(
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int c = g * 8 + j;
-      float m = mean[(long)b * C + c];
-      float rs = rstd[(long)b * C + c];
-      float val = (b2f(v[j]) - m) * rs * gamma[c] + beta[c];
+      float val = (b2f(v[j]) - sm[c]) * sr[c] * gamma[c] + beta[c];
       if (res) val += b2f(rv[j]);
       out[j] = f2b(apply_act(val, act, slope));
     }
@@ -176,29 +180,14 @@ __global__ __launch_bounds__(NT) void in_bwd_reduce_kernel(
   }
 }
 
-// ---- backward: sum the S slabs into s1/s2 [B,C] ----
-__global__ void in_bwd_finalize_kernel(const float* __restrict__ p1,
-                                       const float* __restrict__ p2,
-                                       float* __restrict__ s1,
-                                       float* __restrict__ s2, long n,
-                                       int S) {
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= n) return;
-  float t1 = 0, t2 = 0;
-  for (int k = 0; k < S; ++k) {
-    t1 += p1[(long)k * n + i];
-    t2 += p2[(long)k * n + i];
-  }
-  s1[i] = t1;
-  s2[i] = t2;
-}
+
 
 // ---- backward pass 2: dx; also dgamma/dbeta reduce over b ----
 __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
     const short* __restrict__ dy, const short* __restrict__ x,
     const float* __restrict__ gamma, const float* __restrict__ mean,
-    const float* __restrict__ rstd, const float* __restrict__ s1,
-    const float* __restrict__ s2, short* __restrict__ dx, int B, long HW,
+    const float* __restrict__ rstd, const float* __restrict__ p1,
+    const float* __restrict__ p2, short* __restrict__ dx, int B, long HW,
     int C, int S) {
   int b = blockIdx.x / S;
   int sl = blockIdx.x % S;
@@ -208,6 +197,20 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
   const int tid = threadIdx.x;
   const long base = (long)b * HW * C;
   const float inv_hw = 1.f / (float)HW;
+
+  __shared__ float sm1[MAXC], sm2[MAXC], smean[MAXC], srstd[MAXC];
+  for (int c = tid; c < C; c += NT) {
+    float t1 = 0, t2 = 0;
+    for (int k = 0; k < S; ++k) {
+      t1 += p1[((long)k * B + b) * C + c];
+      t2 += p2[((long)k * B + b) * C + c];
+    }
+    sm1[c] = t1 * inv_hw;
+    sm2[c] = t2 * inv_hw;
+    smean[c] = mean[(long)b * C + c];
+    srstd[c] = rstd[(long)b * C + c];
+  }
+  __syncthreads();
 
   for (long e = tid; e < (r1 - r0) * gpr; e += NT) {
     long r = r0 + e / gpr;
@@ -219,28 +222,25 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int c = g * 8 + j;
-      float m = mean[(long)b * C + c];
-      float rs = rstd[(long)b * C + c];
-      float xh = (b2f(xv[j]) - m) * rs;
+      float rs = srstd[c];
+      float xh = (b2f(xv[j]) - smean[c]) * rs;
       float d = b2f(dv[j]);
-      float m1 = s1[(long)b * C + c] * inv_hw;
-      float m2 = s2[(long)b * C + c] * inv_hw;
-      out[j] = f2b(gamma[c] * rs * (d - m1 - xh * m2));
+      out[j] = f2b(gamma[c] * rs * (d - sm1[c] - xh * sm2[c]));
     }
     *(v8s*)(dx + off) = out;
   }
 }
 
-__global__ void in_bwd_dgb_kernel(const float* __restrict__ s1,
-                                  const float* __restrict__ s2,
+__global__ void in_bwd_dgb_kernel(const float* __restrict__ p1,
+                                  const float* __restrict__ p2,
                                   float* __restrict__ dbeta,
-                                  float* __restrict__ dgamma, int B, int C) {
+                                  float* __restrict__ dgamma, int SB, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float db = 0, dg = 0;
-  for (int b = 0; b < B; ++b) {
-    db += s1[(long)b * C + c];
-    dg += s2[(long)b * C + c];
+  for (int sb = 0; sb < SB; ++sb) {
+    db += p1[(long)sb * C + c];
+    dg += p2[(long)sb * C + c];
   }
   dbeta[c] = db;
   dgamma[c] = dg;
@@ -443,23 +443,18 @@ std::vector<at::Tensor> instnorm_fwd(at::Tensor x, at::Tensor gamma,
                      (const short*)x.const_data_ptr(),
                      (float*)psum.mutable_data_ptr(),
                      (float*)psq.mutable_data_ptr(), B, HW, C, S);
-  long n = (long)B * C;
-  hipLaunchKernelGGL(in_finalize_kernel, dim3(cdiv64(n, 256)), dim3(256), 0,
-                     stream, (const float*)psum.const_data_ptr(),
-                     (const float*)psq.const_data_ptr(),
-                     (float*)mean.mutable_data_ptr(),
-                     (float*)rstd.mutable_data_ptr(), n, S, 1.f / (float)HW,
-                     (float)eps);
   const short* res = residual.has_value()
                          ? (const short*)residual->const_data_ptr() : nullptr;
   hipLaunchKernelGGL(in_norm_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)x.const_data_ptr(),
                      (const float*)gamma.const_data_ptr(),
                      (const float*)beta.const_data_ptr(),
-                     (const float*)mean.const_data_ptr(),
-                     (const float*)rstd.const_data_ptr(), res,
+                     (const float*)psum.const_data_ptr(),
+                     (const float*)psq.const_data_ptr(),
+                     (float*)mean.mutable_data_ptr(),
+                     (float*)rstd.mutable_data_ptr(), res,
                      (short*)y.mutable_data_ptr(), B, HW, C, S, (int)act,
-                     (float)slope);
+                     (float)slope, (float)eps);
   return {y, mean, rstd};
 }
 
@@ -477,8 +472,6 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
   int S = slices_for(HW, B);
   auto p1 = at::empty({S, B, C}, fopt);
   auto p2 = at::empty({S, B, C}, fopt);
-  auto s1 = at::empty({B, C}, fopt);
-  auto s2 = at::empty({B, C}, fopt);
   auto dx = at::empty_like(x);
   auto dgamma = at::empty({C}, fopt);
   auto dbeta = at::empty({C}, fopt);
@@ -489,27 +482,20 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
                      (const float*)rstd.const_data_ptr(),
                      (float*)p1.mutable_data_ptr(),
                      (float*)p2.mutable_data_ptr(), B, HW, C, S);
-  long nbc = (long)B * C;
-  hipLaunchKernelGGL(in_bwd_finalize_kernel, dim3(cdiv64(nbc, 256)),
-                     dim3(256), 0, stream,
-                     (const float*)p1.const_data_ptr(),
-                     (const float*)p2.const_data_ptr(),
-                     (float*)s1.mutable_data_ptr(),
-                     (float*)s2.mutable_data_ptr(), nbc, S);
   hipLaunchKernelGGL(in_bwd_dx_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)dy.const_data_ptr(),
                      (const short*)x.const_data_ptr(),
                      (const float*)gamma.const_data_ptr(),
                      (const float*)mean.const_data_ptr(),
                      (const float*)rstd.const_data_ptr(),
-                     (const float*)s1.const_data_ptr(),
-                     (const float*)s2.const_data_ptr(),
+                     (const float*)p1.const_data_ptr(),
+                     (const float*)p2.const_data_ptr(),
                      (short*)dx.mutable_data_ptr(), B, HW, C, S);
   hipLaunchKernelGGL(in_bwd_dgb_kernel, dim3(cdiv64(C, 256)), dim3(256), 0,
-                     stream, (const float*)s1.const_data_ptr(),
-                     (const float*)s2.const_data_ptr(),
+                     stream, (const float*)p1.const_data_ptr(),
+                     (const float*)p2.const_data_ptr(),
                      (float*)dbeta.mutable_data_ptr(),
-                     (float*)dgamma.mutable_data_ptr(), B, C);
+                     (float*)dgamma.mutable_data_ptr(), S * B, C);
   return {dx, dgamma, dbeta};
 }
 

@@ -221,6 +221,178 @@ __global__ __launch_bounds__(NTHREADS) void conv_gemm_kernel(ConvParams p) {
   }
 }
 
+// ---------------- glds pipelined conv GEMM ----------------
+// Same math as conv_gemm_kernel, but staged by LDS-DMA
+// (buffer_load_dwordx4 ... lds) with double-buffered LDS and an XOR
+// swizzle. The swizzle and the out-of-bounds zero-fill both ride on the
+// per-lane GATHER address: the LDS image is lane-linear (glds requirement),
+// the source lane->element permutation implements byte ^= ((row&7)<<4),
+// and invalid taps (padding, stride phase, M-tail) get an OOB voffset that
+// the buffer descriptor turns into zeros. One __shared__ object only
+// (hipcc de-pipelines glds next to a second one).
+
+struct ConvSmem {
+  short A[2][BM * BK];
+  short Bt[2][BN * BK];
+  long rowyb[BM];
+  int rowih[BM], rowiw[BM];
+  unsigned rowxb[BM];  // byte offset of batch base (tensors < 4 GiB)
+  char rowok[BM];
+};
+
+template <bool IS_CONVT, int STRIDE>
+__global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
+  const int stride = STRIDE ? STRIDE : p.stride;
+  __shared__ ConvSmem sm;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int bid = blockIdx.x;
+  const int mt = bid % p.mtiles, nt = bid / p.mtiles;
+  const long m0 = (long)mt * BM;
+  const int n0 = nt * BN;
+
+  for (int r = tid; r < BM; r += NTHREADS) {
+    long m = m0 + r;
+    bool ok = m < p.M;
+    long mm = ok ? m : 0;
+    int ow = (int)(mm % p.OW);
+    int oh = (int)((mm / p.OW) % p.OH);
+    int b = (int)(mm / ((long)p.OW * p.OH));
+    sm.rowok[r] = ok;
+    sm.rowxb[r] = (unsigned)((long)b * p.H * p.W * p.Cin * 2);
+    sm.rowyb[r] = ((long)(b * p.OH + oh) * p.OW + ow) * p.Cout;
+    if (IS_CONVT) {
+      sm.rowih[r] = oh + p.pt;
+      sm.rowiw[r] = ow + p.pl;
+    } else {
+      sm.rowih[r] = oh * stride - p.pt;
+      sm.rowiw[r] = ow * stride - p.pl;
+    }
+  }
+  __syncthreads();
+
+  // per-lane persistent gather state
+  const int lr = lane >> 3;                    // row-in-8 of each instr
+  const int klog = ((lane & 7) ^ lr) << 3;     // swizzled k-chunk (elems)
+  int aih[4], aiw[4];
+  unsigned axb[4];
+  bool aok[4];
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int r = w * 32 + j * 8 + lr;
+    aih[j] = sm.rowih[r];
+    aiw[j] = sm.rowiw[r];
+    axb[j] = sm.rowxb[r];
+    aok[j] = sm.rowok[r];
+  }
+  const int bn[2] = {(w * 2 + 0) * 8 + lr + n0, (w * 2 + 1) * 8 + lr + n0};
+
+  auto rx = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)p.x, 0, (unsigned)((long)p.B * p.H * p.W * p.Cin * 2), 0x00020000);
+  auto rw = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)p.w, 0, (unsigned)(p.Cout * p.KTOT * 2), 0x00020000);
+
+  const int nk = (int)((p.KTOT + BK - 1) / BK);
+
+  auto stage = [&](int buf, long k0) {
+    long k = k0 + klog;
+    bool kv = k < p.KTOT;
+    int tap = kv ? (int)(k / p.Cin) : 0;
+    int ci = (int)(k - (long)tap * p.Cin);
+    int dkh = tap / p.KW, dkw = tap - (tap / p.KW) * p.KW;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      unsigned vo = 0xFFFFFF00u;
+      if (kv && aok[j]) {
+        bool valid = true;
+        int ih, iw;
+        if (IS_CONVT) {
+          int nh = aih[j] - dkh, nw = aiw[j] - dkw;
+          valid = nh >= 0 && nw >= 0 && (nh % stride) == 0 &&
+                  (nw % stride) == 0;
+          ih = nh / stride; iw = nw / stride;
+          valid = valid && ih < p.H && iw < p.W;
+        } else {
+          ih = aih[j] + dkh; iw = aiw[j] + dkw;
+          if (p.reflect) {
+            ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
+          } else {
+            valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+          }
+        }
+        if (valid)
+          vo = axb[j] + (unsigned)((((long)ih * p.W + iw) * p.Cin + ci) * 2);
+      }
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * 4 + j) * 512],
+          16, vo, 0, 0, 0);
+    }
+    #pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      unsigned vo = 0xFFFFFF00u;
+      if (k < p.KTOT && bn[j] < p.Cout)
+        vo = (unsigned)(((long)bn[j] * p.KTOT + k) * 2);
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rw, (__attribute__((address_space(3))) void*)&sm.Bt[buf][(w * 2 + j) * 512],
+          16, vo, 0, 0, 0);
+    }
+  };
+
+  v4f acc[4][2] = {};
+  const int wr = w >> 1, wc = w & 1;
+  const int wm0 = wr * 64, wn0 = wc * 32;
+  const int fr = lane & 15;
+  const int fg = lane >> 4;
+  const int swz = (fr & 7) << 4;  // read-side XOR (bytes)
+
+  stage(0, 0);
+  __syncthreads();
+
+  for (int kt = 0; kt < nk; ++kt) {
+    if (kt + 1 < nk) stage((kt + 1) & 1, (long)(kt + 1) * BK);
+    const char* Ab = (const char*)sm.A[kt & 1];
+    const char* Bb = (const char*)sm.Bt[kt & 1];
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 32) {
+      const int kbyte = (kk + fg * 8) * 2;
+      v8bf a0 = *(const v8bf*)(Ab + ((wm0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
+      v8bf a1 = *(const v8bf*)(Ab + ((wm0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
+      v8bf a2 = *(const v8bf*)(Ab + ((wm0 + 2 * 16 + fr) << 7) + (kbyte ^ swz));
+      v8bf a3 = *(const v8bf*)(Ab + ((wm0 + 3 * 16 + fr) << 7) + (kbyte ^ swz));
+      v8bf b0 = *(const v8bf*)(Bb + ((wn0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
+      v8bf b1 = *(const v8bf*)(Bb + ((wn0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
+      acc[2][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b0, acc[2][0], 0, 0, 0);
+      acc[2][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b1, acc[2][1], 0, 0, 0);
+      acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b0, acc[3][0], 0, 0, 0);
+      acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b1, acc[3][1], 0, 0, 0);
+    }
+    __syncthreads();  // drains the in-flight glds (vmcnt(0) in the release)
+  }
+
+  #pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+    int n = n0 + wn0 + nf * 16 + fr;
+    if (n >= p.Cout) continue;
+    float bv = p.bias ? b2f(p.bias[n]) : 0.f;
+    #pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int rl = wm0 + mf * 16 + fg * 4 + r;
+        if (!sm.rowok[rl]) continue;
+        float v = apply_act(acc[mf][nf][r] + bv, p.act, p.slope);
+        p.y[sm.rowyb[rl] + n] = f2b(v);
+      }
+    }
+  }
+}
+
 // ---------------- weight gradient ----------------
 // dw[n][k] = Σ_m A[m][k] · dy[m][n];  A-tile and dy-tile staged TRANSPOSED
 // (m contiguous per row) so the MFMA reduce dim is m. fp32 atomics over
@@ -464,9 +636,35 @@ static void launch_conv_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
   }
 }
 
+template <bool IS_CONVT>
+static void launch_glds_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
+  switch (p.stride) {
+    case 1:
+      hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 1>), grid,
+                         dim3(NTHREADS), 0, stream, p);
+      break;
+    case 2:
+      hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 2>), grid,
+                         dim3(NTHREADS), 0, stream, p);
+      break;
+    default:
+      hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 0>), grid,
+                         dim3(NTHREADS), 0, stream, p);
+  }
+}
+
 static void launch_conv(const ConvParams& p, bool is_convt, hipStream_t stream) {
   dim3 grid(p.mtiles * p.ntiles);
   bool aligned = (p.Cin % 8) == 0;
+  // glds path needs 32-bit byte offsets into x and w
+  bool glds_ok = aligned &&
+                 (long)p.B * p.H * p.W * p.Cin * 2 < (1L << 31) &&
+                 (long)p.Cout * p.KTOT * 2 < (1L << 31);
+  if (glds_ok) {
+    if (is_convt) launch_glds_s<true>(p, grid, stream);
+    else launch_glds_s<false>(p, grid, stream);
+    return;
+  }
   if (is_convt) {
     if (aligned) launch_conv_s<true, true>(p, grid, stream);
     else launch_conv_s<true, false>(p, grid, stream);
@@ -603,6 +801,37 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
   hipLaunchKernelGGL(wgrad_kernel, grid, dim3(NTHREADS), 0,
                      at::cuda::getCurrentCUDAStream(), p);
   return dw;
+}
+
+// ---- glds probe (test aid): verify raw_ptr_buffer_load_lds semantics ----
+// lane l loads 16B from x at voff[l]; expect LDS dest = base + l*16 and
+// OOB voffsets to produce zeros.
+__global__ void glds_probe_kernel(const float* x, const unsigned* voff,
+                                  int nbytes, float* out) {
+  __shared__ float lds[128];
+  int l = threadIdx.x;  // 64 threads
+  lds[l] = -1.f; lds[64 + l] = -1.f;
+  __syncthreads();
+  auto rsrc = __builtin_amdgcn_make_buffer_rsrc((void*)x, 0, nbytes, 0x00020000);
+  __builtin_amdgcn_raw_ptr_buffer_load_lds(
+      rsrc, (__attribute__((address_space(3))) void*)&lds[0], 16, voff[l], 0,
+      0, 0);
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+  out[l] = lds[l];
+  out[64 + l] = lds[64 + l];
+}
+
+at::Tensor glds_probe(at::Tensor x, at::Tensor voff) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat);
+  TORCH_CHECK(voff.scalar_type() == at::kInt && voff.numel() == 64);
+  auto out = at::empty({128}, x.options());
+  hipLaunchKernelGGL(glds_probe_kernel, dim3(1), dim3(64), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     (const float*)x.const_data_ptr(),
+                     (const unsigned*)voff.const_data_ptr(),
+                     (int)(x.numel() * 4), (float*)out.mutable_data_ptr());
+  return out;
 }
 
 // ---- MFMA layout probe (test aid): C[16,16] = A[16,32] @ Bt[16,32]^T ----

@@ -417,6 +417,10 @@ constexpr int WG_LDM = WG_BM + 8;
 __global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
   __shared__ short At[WG_BK][WG_LDM];   // [k][m]
   __shared__ short Dt[WG_BN][WG_LDM];   // [n][m]
+  // per-m gather state, decoded once per m-step by 64 threads (32-bit ops)
+  __shared__ int mih[WG_BM], miw[WG_BM];
+  __shared__ long mxb[WG_BM];
+  __shared__ char mok[WG_BM];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -443,34 +447,54 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
 
   for (long ms = mstart; ms < mend; ms += WG_BM) {
     __syncthreads();  // previous iteration's MFMA reads done
-    // ---- stage At[k][m]: chunk = (m, 8 k-rows) ----
+    // decode the 64 m-coordinates once (32-bit divisions)
+    for (int r = tid; r < WG_BM; r += NTHREADS) {
+      int m = (int)(ms + r);
+      bool ok = ms + r < p.M;
+      int mm = ok ? m : 0;
+      int ow = mm % p.OW;
+      int t = mm / p.OW;
+      int oh = t % p.OH;
+      int b = t / p.OH;
+      mok[r] = ok;
+      mih[r] = oh * p.stride - p.pt;
+      miw[r] = ow * p.stride - p.pl;
+      mxb[r] = (long)b * p.H * p.W * p.Cin;
+    }
+    __syncthreads();
+    // ---- stage At[k][m]: chunk = (4 m, 8 k-rows); register transpose,
+    // 4x v8 global loads + 8x ds_write_b64 (conflict-free) ----
     if (a_vec) {
-      for (int c = tid; c < WG_BM * (WG_BK / 8); c += NTHREADS) {
-        int m_loc = c % WG_BM;
-        int kc = (c / WG_BM) * 8;      // first of 8 consecutive k (same tap)
+      for (int c = tid; c < (WG_BM / 4) * (WG_BK / 8); c += NTHREADS) {
+        int m_loc = (c % (WG_BM / 4)) * 4;
+        int kc = (c / (WG_BM / 4)) * 8;  // first of 8 consecutive k (same tap)
         long k = k0 + kc;
-        v8s val = {};
-        long m = ms + m_loc;
-        if (k < p.KTOT && m < p.M) {
+        v8s vals[4] = {{}, {}, {}, {}};
+        if (k < p.KTOT) {
           int tap = (int)(k / p.Cin);
           int ci = (int)(k - (long)tap * p.Cin);
           int dkh = tap / p.KW, dkw = tap - (tap / p.KW) * p.KW;
-          int ow = (int)(m % p.OW);
-          int oh = (int)((m / p.OW) % p.OH);
-          int b = (int)(m / ((long)p.OW * p.OH));
-          int ih = oh * p.stride - p.pt + dkh;
-          int iw = ow * p.stride - p.pl + dkw;
-          bool valid = true;
-          if (p.reflect) {
-            ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
-          } else {
-            valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+          #pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            if (!mok[m_loc + u]) continue;
+            int ih = mih[m_loc + u] + dkh;
+            int iw = miw[m_loc + u] + dkw;
+            bool valid = true;
+            if (p.reflect) {
+              ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
+            } else {
+              valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+            }
+            if (valid)
+              vals[u] = *(const v8s*)(p.x + mxb[m_loc + u] +
+                                      ((long)ih * p.W + iw) * p.Cin + ci);
           }
-          if (valid)
-            val = *(const v8s*)(p.x + (((long)b * p.H + ih) * p.W + iw) * p.Cin + ci);
         }
         #pragma unroll
-        for (int j = 0; j < 8; ++j) At[kc + j][m_loc] = val[j];
+        for (int j = 0; j < 8; ++j) {
+          short pack[4] = {vals[0][j], vals[1][j], vals[2][j], vals[3][j]};
+          *(uint2*)&At[kc + j][m_loc] = *(uint2*)pack;
+        }
       }
     } else {
       for (int c = tid; c < WG_BK * 8; c += NTHREADS) {
@@ -509,17 +533,22 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
         *(v8s*)&At[krow][mc] = *(v8s*)vals;
       }
     }
-    // ---- stage Dt[n][m]: chunk = (m, 8 n) ----
+    // ---- stage Dt[n][m]: chunk = (2 m, 8 n); register transpose ----
     if (d_vec) {
-      for (int c = tid; c < WG_BM * (WG_BN / 8); c += NTHREADS) {
-        int m_loc = c % WG_BM;
-        int nc = (c / WG_BM) * 8;
-        long m = ms + m_loc;
-        v8s val = {};
-        if (m < p.M && n0 + nc < p.Cout)
-          val = *(const v8s*)(p.dy + m * p.Cout + n0 + nc);
+      for (int c = tid; c < (WG_BM / 2) * (WG_BN / 8); c += NTHREADS) {
+        int m_loc = (c % (WG_BM / 2)) * 2;
+        int nc = (c / (WG_BM / 2)) * 8;
+        v8s v0 = {}, v1 = {};
+        if (n0 + nc < p.Cout) {
+          long m = ms + m_loc;
+          if (m < p.M) v0 = *(const v8s*)(p.dy + m * p.Cout + n0 + nc);
+          if (m + 1 < p.M) v1 = *(const v8s*)(p.dy + (m + 1) * p.Cout + n0 + nc);
+        }
         #pragma unroll
-        for (int j = 0; j < 8; ++j) Dt[nc + j][m_loc] = val[j];
+        for (int j = 0; j < 8; ++j) {
+          short pack[2] = {v0[j], v1[j]};
+          *(unsigned*)&Dt[nc + j][m_loc] = *(unsigned*)pack;
+        }
       }
     } else {
       for (int c = tid; c < WG_BN * 8; c += NTHREADS) {

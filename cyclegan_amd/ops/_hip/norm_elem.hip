@@ -230,19 +230,22 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
   }
 }
 
+// grid (cdiv(C,256), nchunks): each block sums 8 sb-rows, atomics once.
 __global__ void in_bwd_dgb_kernel(const float* __restrict__ p1,
                                   const float* __restrict__ p2,
                                   float* __restrict__ dbeta,
                                   float* __restrict__ dgamma, int SB, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
+  int sb0 = blockIdx.y * 8;
+  int sb1 = min(sb0 + 8, SB);
   float db = 0, dg = 0;
-  for (int sb = 0; sb < SB; ++sb) {
+  for (int sb = sb0; sb < sb1; ++sb) {
     db += p1[(long)sb * C + c];
     dg += p2[(long)sb * C + c];
   }
-  dbeta[c] = db;
-  dgamma[c] = dg;
+  atomicAdd(&dbeta[c], db);
+  atomicAdd(&dgamma[c], dg);
 }
 
 // ---- activation backward (from output) ----
@@ -472,8 +475,8 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
   auto p1 = at::empty({S, B, C}, fopt);
   auto p2 = at::empty({S, B, C}, fopt);
   auto dx = at::empty_like(x);
-  auto dgamma = at::empty({C}, fopt);
-  auto dbeta = at::empty({C}, fopt);
+  auto dgamma = at::zeros({C}, fopt);
+  auto dbeta = at::zeros({C}, fopt);
   hipLaunchKernelGGL(in_bwd_reduce_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)dy.const_data_ptr(),
                      (const short*)x.const_data_ptr(),
@@ -490,7 +493,8 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
                      (const float*)p1.const_data_ptr(),
                      (const float*)p2.const_data_ptr(),
                      (short*)dx.mutable_data_ptr(), B, HW, C, S);
-  hipLaunchKernelGGL(in_bwd_dgb_kernel, dim3(cdiv64(C, 256)), dim3(256), 0,
+  hipLaunchKernelGGL(in_bwd_dgb_kernel,
+                     dim3(cdiv64(C, 256), cdiv64(S * B, 8)), dim3(256), 0,
                      stream, (const float*)p1.const_data_ptr(),
                      (const float*)p2.const_data_ptr(),
                      (float*)dbeta.mutable_data_ptr(),

@@ -16,6 +16,7 @@ at::Tensor conv2d_wgrad(at::Tensor, at::Tensor, int64_t, int64_t, int64_t,
                         int64_t, int64_t, bool);
 at::Tensor mfma_probe(at::Tensor, at::Tensor);
 at::Tensor glds_probe(at::Tensor, at::Tensor);
+at::Tensor tr_probe(int64_t);
 std::vector<at::Tensor> instnorm_fwd(at::Tensor, at::Tensor, at::Tensor,
                                      double, int64_t, double,
                                      c10::optional<at::Tensor>);
@@ -41,6 +42,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_wgrad", &cyg::conv2d_wgrad);
   m.def("mfma_probe", &cyg::mfma_probe);
   m.def("glds_probe", &cyg::glds_probe);
+  m.def("tr_probe", &cyg::tr_probe);
   m.def("instnorm_fwd", &cyg::instnorm_fwd);
   m.def("instnorm_bwd", &cyg::instnorm_bwd);
   m.def("act_bwd", &cyg::act_bwd);

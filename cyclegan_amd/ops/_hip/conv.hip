@@ -417,10 +417,10 @@ constexpr int WG_LDM = WG_BM + 8;
 __global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
   __shared__ short At[WG_BK][WG_LDM];   // [k][m]
   __shared__ short Dt[WG_BN][WG_LDM];   // [n][m]
-  // per-m gather state, decoded once per m-step by 64 threads (32-bit ops)
-  __shared__ int mih[WG_BM], miw[WG_BM];
-  __shared__ long mxb[WG_BM];
-  __shared__ char mok[WG_BM];
+  // per-m gather state, double-buffered, decoded once per m-step
+  __shared__ int mih[2][WG_BM], miw[2][WG_BM];
+  __shared__ long mxb[2][WG_BM];
+  __shared__ char mok[2][WG_BM];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -440,10 +440,121 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
   const int fr = lane & 15, fg = lane >> 4;
 
   // Staging is transposed: global loads VECTORIZE along the natural inner
-  // dims (ci for x, cout for dy) and scatter 8 scalar ds_writes into the
-  // [k][m] / [n][m] images (LDS scatter ≪ scalar global gather).
+  // dims (ci for x, cout for dy) and scatter into the [k][m] / [n][m]
+  // images as b64/b32 writes. The vectorized path register-pipelines the
+  // next m-step's loads under the current step's MFMAs (T14 split).
   const bool a_vec = (p.Cin % 8) == 0;
   const bool d_vec = (p.Cout % 8) == 0;
+
+  if (a_vec && d_vec && mstart < mend) {
+    // fixed per-thread chunk identities (exactly 256 chunks each)
+    const int a_mloc = (tid & 15) * 4, a_kc = (tid >> 4) * 8;
+    const int d_mloc = (tid & 31) * 2, d_nc = (tid >> 5) * 8;
+    const long ak = k0 + a_kc;
+    const bool akv = ak < p.KTOT;
+    int tap = akv ? (int)(ak / p.Cin) : 0;
+    const int ci = (int)(ak - (long)tap * p.Cin);
+    const int dkh = tap / p.KW, dkw = tap - (tap / p.KW) * p.KW;
+    const bool dnv = n0 + d_nc < p.Cout;
+
+    auto decode = [&](int buf, long ms) {
+      for (int r = tid; r < WG_BM; r += NTHREADS) {
+        bool ok = ms + r < p.M;
+        int mm = ok ? (int)(ms + r) : 0;
+        int ow = mm % p.OW;
+        int t = mm / p.OW;
+        int oh = t % p.OH;
+        int b = t / p.OH;
+        mok[buf][r] = ok;
+        mih[buf][r] = oh * p.stride - p.pt;
+        miw[buf][r] = ow * p.stride - p.pl;
+        mxb[buf][r] = (long)b * p.H * p.W * p.Cin;
+      }
+    };
+
+    const v8s VZERO = {};
+    v8s areg[4], dreg[2];
+    auto load_regs = [&](int buf, long ms) {
+      #pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        areg[u] = VZERO;
+        if (akv && mok[buf][a_mloc + u]) {
+          int ih = mih[buf][a_mloc + u] + dkh;
+          int iw = miw[buf][a_mloc + u] + dkw;
+          bool valid = true;
+          if (p.reflect) {
+            ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
+          } else {
+            valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+          }
+          if (valid)
+            areg[u] = *(const v8s*)(p.x + mxb[buf][a_mloc + u] +
+                                    ((long)ih * p.W + iw) * p.Cin + ci);
+        }
+      }
+      #pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        dreg[u] = VZERO;
+        long m = ms + d_mloc + u;
+        if (dnv && m < p.M)
+          dreg[u] = *(const v8s*)(p.dy + m * p.Cout + n0 + d_nc);
+      }
+    };
+
+    decode(0, mstart);
+    __syncthreads();
+    load_regs(0, mstart);
+
+    int cur = 0;
+    for (long ms = mstart; ms < mend; ms += WG_BM, cur ^= 1) {
+      // write the registers staged last iteration into the LDS tiles
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        short pa[4] = {areg[0][j], areg[1][j], areg[2][j], areg[3][j]};
+        *(uint2*)&At[a_kc + j][a_mloc] = *(uint2*)pa;
+        short pd[2] = {dreg[0][j], dreg[1][j]};
+        *(unsigned*)&Dt[d_nc + j][d_mloc] = *(unsigned*)pd;
+      }
+      bool more = ms + WG_BM < mend;
+      if (more) decode(cur ^ 1, ms + WG_BM);
+      __syncthreads();
+      if (more) load_regs(cur ^ 1, ms + WG_BM);  // overlaps the MFMAs below
+      #pragma unroll
+      for (int kk = 0; kk < WG_BM; kk += 32) {
+        v8bf a0 = *(const v8bf*)&At[wk0 + 0 * 16 + fr][kk + fg * 8];
+        v8bf a1 = *(const v8bf*)&At[wk0 + 1 * 16 + fr][kk + fg * 8];
+        v8bf a2 = *(const v8bf*)&At[wk0 + 2 * 16 + fr][kk + fg * 8];
+        v8bf a3 = *(const v8bf*)&At[wk0 + 3 * 16 + fr][kk + fg * 8];
+        v8bf b0 = *(const v8bf*)&Dt[wn0 + 0 * 16 + fr][kk + fg * 8];
+        v8bf b1 = *(const v8bf*)&Dt[wn0 + 1 * 16 + fr][kk + fg * 8];
+        acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
+        acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
+        acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
+        acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
+        acc[2][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b0, acc[2][0], 0, 0, 0);
+        acc[2][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b1, acc[2][1], 0, 0, 0);
+        acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b0, acc[3][0], 0, 0, 0);
+        acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b1, acc[3][1], 0, 0, 0);
+      }
+      __syncthreads();
+    }
+
+    #pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      int n = n0 + wn0 + nf * 16 + fr;
+      if (n >= p.Cout) continue;
+      #pragma unroll
+      for (int kf = 0; kf < 4; ++kf) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          long k = k0 + wk0 + kf * 16 + fg * 4 + r;
+          if (k < p.KTOT)
+            atomicAdd(&p.dw[(long)n * p.KTOT + k], acc[kf][nf][r]);
+        }
+      }
+    }
+    return;
+  }
 
   for (long ms = mstart; ms < mend; ms += WG_BM) {
     __syncthreads();  // previous iteration's MFMA reads done
@@ -456,47 +567,13 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
       int t = mm / p.OW;
       int oh = t % p.OH;
       int b = t / p.OH;
-      mok[r] = ok;
-      mih[r] = oh * p.stride - p.pt;
-      miw[r] = ow * p.stride - p.pl;
-      mxb[r] = (long)b * p.H * p.W * p.Cin;
+      mok[0][r] = ok;
+      mih[0][r] = oh * p.stride - p.pt;
+      miw[0][r] = ow * p.stride - p.pl;
+      mxb[0][r] = (long)b * p.H * p.W * p.Cin;
     }
     __syncthreads();
-    // ---- stage At[k][m]: chunk = (4 m, 8 k-rows); register transpose,
-    // 4x v8 global loads + 8x ds_write_b64 (conflict-free) ----
-    if (a_vec) {
-      for (int c = tid; c < (WG_BM / 4) * (WG_BK / 8); c += NTHREADS) {
-        int m_loc = (c % (WG_BM / 4)) * 4;
-        int kc = (c / (WG_BM / 4)) * 8;  // first of 8 consecutive k (same tap)
-        long k = k0 + kc;
-        v8s vals[4] = {{}, {}, {}, {}};
-        if (k < p.KTOT) {
-          int tap = (int)(k / p.Cin);
-          int ci = (int)(k - (long)tap * p.Cin);
-          int dkh = tap / p.KW, dkw = tap - (tap / p.KW) * p.KW;
-          #pragma unroll
-          for (int u = 0; u < 4; ++u) {
-            if (!mok[m_loc + u]) continue;
-            int ih = mih[m_loc + u] + dkh;
-            int iw = miw[m_loc + u] + dkw;
-            bool valid = true;
-            if (p.reflect) {
-              ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
-            } else {
-              valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
-            }
-            if (valid)
-              vals[u] = *(const v8s*)(p.x + mxb[m_loc + u] +
-                                      ((long)ih * p.W + iw) * p.Cin + ci);
-          }
-        }
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          short pack[4] = {vals[0][j], vals[1][j], vals[2][j], vals[3][j]};
-          *(uint2*)&At[kc + j][m_loc] = *(uint2*)pack;
-        }
-      }
-    } else {
+    {
       for (int c = tid; c < WG_BK * 8; c += NTHREADS) {
         int krow = c >> 3;
         int mc = (c & 7) * 8;
@@ -587,6 +664,183 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
   }
 
   // ---- atomic accumulate: D row = k (= (lane>>4)*4+r), col = n ----
+  #pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+    int n = n0 + wn0 + nf * 16 + fr;
+    if (n >= p.Cout) continue;
+    #pragma unroll
+    for (int kf = 0; kf < 4; ++kf) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long k = k0 + wk0 + kf * 16 + fg * 4 + r;
+        if (k < p.KTOT)
+          atomicAdd(&p.dw[(long)n * p.KTOT + k], acc[kf][nf][r]);
+      }
+    }
+  }
+}
+
+// ---------------- glds + tr_b16 weight gradient ----------------
+// Natural [m][k] / [m][n] LDS images staged by LDS-DMA (the same coalesced
+// gather as the fwd kernel), consumed TRANSPOSED by ds_read_b64_tr_b16:
+// per 16-lane group the 16 8-byte loads form a [4][16] bf16 matrix and
+// lane l receives column l&15 (verified by the tr_probe binding), so the
+// MFMA reduce dim is m with zero shuffle cost. Double-buffered, XOR
+// swizzle rides on the gather lane->element assignment:
+//   A image: [64 m][256 B], stored cb = logical_cb ^ ((m&7)<<5)
+//   D image: [64 m][128 B], stored cb = logical_cb ^ ((m&3)<<5)
+
+typedef bf16r v4bfx __attribute__((ext_vector_type(4)));
+
+DEV v4bfx tr16_read(const char* plds) {
+  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) v4bfx*)plds);
+}
+
+struct WgSmem {
+  short A[2][WG_BM * WG_BK];   // [m][k] swizzled
+  short D[2][WG_BM * WG_BN];   // [m][n] swizzled
+};
+
+__global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
+  __shared__ WgSmem sm;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int kt = blockIdx.x % p.ktiles;
+  const int nt = (blockIdx.x / p.ktiles) % p.ntiles;
+  const int sl = blockIdx.x / (p.ktiles * p.ntiles);
+  const long k0 = (long)kt * WG_BK;
+  const int n0 = nt * WG_BN;
+  const long mstart = sl * p.mchunks_per_slice * WG_BM;
+  long mend = mstart + p.mchunks_per_slice * WG_BM;
+  if (mend > p.M) mend = p.M;
+  if (mstart >= mend) return;
+
+  auto rx = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)p.x, 0, (unsigned)((long)p.B * p.H * p.W * p.Cin * 2), 0x00020000);
+  auto rd = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)p.dy, 0, (unsigned)(p.M * p.Cout * 2), 0x00020000);
+
+  // ---- per-lane gather identities ----
+  // A: 16 instrs (4/wave): instr j: row = (w*4+j)*4 + lane/16,
+  //    cb_lin = (lane%16)*16, logical k = (cb_lin ^ ((row&7)<<5))/2 + k0
+  int a_row[4], a_k[4], a_tap[4], a_ci[4], a_dkh[4], a_dkw[4];
+  bool a_kv[4];
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int row = (w * 4 + j) * 4 + (lane >> 4);
+    int cb = ((lane & 15) * 16) ^ ((row & 7) << 5);
+    long k = k0 + cb / 2;
+    a_row[j] = row;
+    a_kv[j] = k < p.KTOT;
+    int tap = a_kv[j] ? (int)(k / p.Cin) : 0;
+    a_ci[j] = (int)(k - (long)tap * p.Cin);
+    a_dkh[j] = tap / p.KW;
+    a_dkw[j] = tap - a_dkh[j] * p.KW;
+    a_k[j] = (int)(k - k0);
+  }
+  // D: 8 instrs (2/wave): instr j: row = (w*2+j)*8 + lane/8,
+  //    cb_lin = (lane%8)*16, logical n = (cb_lin ^ ((row&3)<<5))/2 + n0
+  int d_row[2], d_n[2];
+  #pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    int row = (w * 2 + j) * 8 + (lane >> 3);
+    int cb = ((lane & 7) * 16) ^ ((row & 3) << 5);
+    d_row[j] = row;
+    d_n[j] = n0 + cb / 2;
+  }
+
+  auto stage = [&](int buf, long ms) {
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      unsigned vo = 0xFFFFFF00u;
+      long m = ms + a_row[j];
+      if (a_kv[j] && m < p.M) {
+        int ow = (int)(m % p.OW);
+        int t = (int)(m / p.OW);
+        int oh = t % p.OH;
+        int b = t / p.OH;
+        int ih = oh * p.stride - p.pt + a_dkh[j];
+        int iw = ow * p.stride - p.pl + a_dkw[j];
+        bool valid = true;
+        if (p.reflect) {
+          ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
+        } else {
+          valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+        }
+        if (valid)
+          vo = (unsigned)(((((long)b * p.H + ih) * p.W + iw) * p.Cin + a_ci[j]) * 2);
+      }
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * 4 + j) * 512],
+          16, vo, 0, 0, 0);
+    }
+    #pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      unsigned vo = 0xFFFFFF00u;
+      long m = ms + d_row[j];
+      if (m < p.M && d_n[j] < p.Cout)
+        vo = (unsigned)((m * p.Cout + d_n[j]) * 2);
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rd, (__attribute__((address_space(3))) void*)&sm.D[buf][(w * 2 + j) * 512],
+          16, vo, 0, 0, 0);
+    }
+  };
+
+  v4f acc[4][2] = {};
+  const int wr = w >> 1, wc = w & 1;
+  const int wk0 = wr * 64, wn0 = wc * 32;  // wave tile: 64 k x 32 n
+  const int fr = lane & 15, fg = lane >> 4;
+  const int jg = lane & 15;
+  // tr-read per-lane address components (bytes)
+  const int tr_row_a = jg >> 2;            // row-in-4 within the tr tile
+  const int tr_cb_a = (jg & 3) * 8;        // 4-elem column sub-offset
+
+  stage(0, mstart);
+  __syncthreads();
+
+  int cur = 0;
+  for (long ms = mstart; ms < mend; ms += WG_BM, cur ^= 1) {
+    if (ms + WG_BM < mend) stage(cur ^ 1, ms + WG_BM);
+    const char* Ab = (const char*)sm.A[cur];
+    const char* Db = (const char*)sm.D[cur];
+    #pragma unroll
+    for (int kk = 0; kk < WG_BM; kk += 32) {
+      // A fragments: MFMA row = weight-k col0+fr, reduce elems = m
+      v8bf a[4];
+      #pragma unroll
+      for (int kf = 0; kf < 4; ++kf) {
+        int cbl = (wk0 + kf * 16) * 2 + tr_cb_a;
+        int row = kk + fg * 8 + tr_row_a;
+        v4bfx lo = tr16_read(Ab + row * 256 + (cbl ^ ((row & 7) << 5)));
+        row += 4;
+        v4bfx hi = tr16_read(Ab + row * 256 + (cbl ^ ((row & 7) << 5)));
+        a[kf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+      v8bf bfr[2];
+      #pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int cbl = (wn0 + nf * 16) * 2 + tr_cb_a;
+        int row = kk + fg * 8 + tr_row_a;
+        v4bfx lo = tr16_read(Db + row * 128 + (cbl ^ ((row & 3) << 5)));
+        row += 4;
+        v4bfx hi = tr16_read(Db + row * 128 + (cbl ^ ((row & 3) << 5)));
+        bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[0], bfr[0], acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[0], bfr[1], acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[1], bfr[0], acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[1], bfr[1], acc[1][1], 0, 0, 0);
+      acc[2][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[2], bfr[0], acc[2][0], 0, 0, 0);
+      acc[2][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[2], bfr[1], acc[2][1], 0, 0, 0);
+      acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[3], bfr[0], acc[3][0], 0, 0, 0);
+      acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[3], bfr[1], acc[3][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
   #pragma unroll
   for (int nf = 0; nf < 2; ++nf) {
     int n = n0 + wn0 + nf * 16 + fr;
@@ -827,8 +1081,15 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
                       x.options().dtype(at::kFloat));
   p.dw = (float*)dw.mutable_data_ptr();
   dim3 grid((long)p.ktiles * p.ntiles * p.slices);
-  hipLaunchKernelGGL(wgrad_kernel, grid, dim3(NTHREADS), 0,
-                     at::cuda::getCurrentCUDAStream(), p);
+  bool glds_ok = (p.Cin % 8) == 0 && (p.Cout % 8) == 0 &&
+                 (long)p.B * p.H * p.W * p.Cin * 2 < (1L << 31) &&
+                 p.M * p.Cout * 2 < (1L << 31);
+  if (glds_ok)
+    hipLaunchKernelGGL(wgrad_glds_kernel, grid, dim3(NTHREADS), 0,
+                       at::cuda::getCurrentCUDAStream(), p);
+  else
+    hipLaunchKernelGGL(wgrad_kernel, grid, dim3(NTHREADS), 0,
+                       at::cuda::getCurrentCUDAStream(), p);
   return dw;
 }
 
@@ -860,6 +1121,37 @@ at::Tensor glds_probe(at::Tensor x, at::Tensor voff) {
                      (const float*)x.const_data_ptr(),
                      (const unsigned*)voff.const_data_ptr(),
                      (int)(x.numel() * 4), (float*)out.mutable_data_ptr());
+  return out;
+}
+
+// ---- ds_read_b64_tr_b16 probe: identity-filled LDS, per-lane address by
+// mode; out[l*4+j] = LDS element index that lane l's element j received.
+typedef bf16r v4bf __attribute__((ext_vector_type(4)));
+typedef short v4sh __attribute__((ext_vector_type(4)));
+__global__ void tr_probe_kernel(short* out, int mode) {
+  __shared__ short lds[1024];
+  int l = threadIdx.x;
+  for (int i = l; i < 1024; i += 64) lds[i] = (short)i;
+  __syncthreads();
+  int addr;
+  switch (mode) {
+    case 0: addr = (l & 15) + (l >> 4) * 64; break;
+    case 1: addr = 0; break;
+    case 2: addr = l * 4; break;
+    default: addr = (l & 15) * 2 + (l >> 4) * 64; break;
+  }
+  auto p = (__attribute__((address_space(3))) v4bf*)(
+      (__attribute__((address_space(3))) short*)lds + addr);
+  v4bf r = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) out[l * 4 + j] = ((v4sh&)r)[j];
+}
+
+at::Tensor tr_probe(int64_t mode) {
+  auto out = at::empty({256}, at::TensorOptions().dtype(at::kShort).device(at::kCUDA));
+  hipLaunchKernelGGL(tr_probe_kernel, dim3(1), dim3(64), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     (short*)out.mutable_data_ptr(), (int)mode);
   return out;
 }
 

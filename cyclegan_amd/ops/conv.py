@@ -33,7 +33,8 @@ import torch
 import torch.nn.functional as F
 
 from . import backend
-from .shadow import compute_weight, compute_weight_t
+from .shadow import (compute_weight, compute_weight_t, compute_weight_p,
+                     compute_weight_tp)
 
 ACT_NONE, ACT_RELU, ACT_LRELU, ACT_TANH = 0, 1, 2, 3
 
@@ -93,40 +94,67 @@ def _convt_ref(x, w, bias, stride, pt, pl, out_h, out_w):
     return y.permute(0, 2, 3, 1)
 
 
+def _pad_channels(t: torch.Tensor, c_to: int = 8) -> torch.Tensor:
+    """Zero-pad the channel (last) dim of an NHWC tensor up to c_to."""
+    c = t.shape[-1]
+    if c >= c_to:
+        return t
+    return F.pad(t, (0, c_to - c))
+
+
 class _ConvFn(torch.autograd.Function):
-    """HIP implicit-GEMM conv with fused pad + bias + activation."""
+    """HIP implicit-GEMM conv with fused pad + bias + activation.
+
+    Channel dims < 8 (the RGB stem / 1-3 channel heads) are zero-padded to
+    8 so every shape runs the fast glds kernels; padded input channels see
+    zero activations (grads exactly zero) and padded output channels are
+    sliced off."""
 
     @staticmethod
     def forward(ctx, x, w, bias, stride, pads, reflect, act, slope):
         ext = backend.ext()
-        wc = compute_weight(w, x)
+        cout = w.shape[0]
+        xp = _pad_channels(x).contiguous()
+        wc = compute_weight_p(w, x)
         bc = compute_weight(bias, x) if bias is not None else None
-        y = ext.conv2d_fwd(x, wc, bc, stride, *pads, reflect, act, slope)
-        ctx.save_for_backward(x, w, y)
-        ctx.conf = (stride, pads, reflect, act, slope, bias is not None)
+        if bc is not None and cout < 8:
+            bc = _pad_channels(bc.view(1, 1, 1, -1)).view(-1)
+        y = ext.conv2d_fwd(xp, wc, bc, stride, *pads, reflect, act, slope)
+        if cout < 8:
+            y = y[..., :cout].contiguous()
+        ctx.save_for_backward(xp, w, y)
+        ctx.conf = (stride, pads, reflect, act, slope, bias is not None,
+                    x.shape[3])
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, w, y = ctx.saved_tensors
-        stride, pads, reflect, act, slope, has_bias = ctx.conf
+        xp, w, y = ctx.saved_tensors
+        stride, pads, reflect, act, slope, has_bias, cin = ctx.conf
+        cout = w.shape[0]
         pt, pb, pl, pr = pads
         ext = backend.ext()
         dy = dy.contiguous()
         if act != ACT_NONE:
             dy = ext.act_bwd(dy, y, act, slope)
-        dx = dw = db = None
-        if ctx.needs_input_grad[0]:
-            wt = compute_weight_t(w, x)
-            dx = ext.conv2d_dgrad(dy, wt, x.shape[1], x.shape[2], stride,
-                                  pt, pb, pl, pr, reflect)
-        if ctx.needs_input_grad[1]:
-            dw = ext.conv2d_wgrad(x, dy, w.shape[1], w.shape[2], stride,
-                                  pt, pl, reflect)
-            if dw.dtype != w.dtype:
-                dw = dw.to(w.dtype)
+        db = None
         if has_bias and ctx.needs_input_grad[2]:
             db = dy.float().sum(dim=(0, 1, 2))
+        dyp = _pad_channels(dy).contiguous()
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            wt = compute_weight_tp(w, xp)
+            dx = ext.conv2d_dgrad(dyp, wt, xp.shape[1], xp.shape[2], stride,
+                                  pt, pb, pl, pr, reflect)
+            if cin < 8:
+                dx = dx[..., :cin].contiguous()
+        if ctx.needs_input_grad[1]:
+            dw = ext.conv2d_wgrad(xp, dyp, w.shape[1], w.shape[2], stride,
+                                  pt, pl, reflect)
+            if cout < 8 or cin < 8:
+                dw = dw[:cout, :, :, :cin].contiguous()
+            if dw.dtype != w.dtype:
+                dw = dw.to(w.dtype)
         return dx, dw, db, None, None, None, None, None
 
 

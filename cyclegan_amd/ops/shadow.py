@@ -15,6 +15,8 @@ from torch.utils.weak import WeakTensorKeyDictionary
 
 _cache: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
 _cache_t: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
+_cache_p: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
+_cache_tp: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
 
 
 def bf16_shadow(t: torch.Tensor) -> torch.Tensor:
@@ -37,6 +39,51 @@ def compute_weight(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
     if w.dtype != like.dtype:
         return w.detach().to(like.dtype)
     return w.detach()
+
+
+def _pad_dims(w: torch.Tensor) -> torch.Tensor:
+    """Zero-pad OHWI weight dims O and I up to 8 (glds kernel alignment).
+    Padded input channels see zero activations and padded output channels
+    are sliced away, so the math is unchanged."""
+    cout, kh, kw, cin = w.shape
+    po = max(0, 8 - cout)
+    pi = max(0, 8 - cin)
+    if po == 0 and pi == 0:
+        return w
+    return torch.nn.functional.pad(w, (0, pi, 0, 0, 0, 0, 0, po))
+
+
+def compute_weight_p(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
+    """Padded (dims >= 8) compute-dtype weight, cached per master version."""
+    ent = _cache_p.get(w)
+    ver = w._version
+    key = (ver, like.dtype, "p")
+    if ent is not None and ent[0] == key:
+        return ent[1]
+    wp = _pad_dims(w.detach())
+    if like.dtype == torch.bfloat16 and wp.dtype != torch.bfloat16:
+        wp = wp.to(torch.bfloat16)
+    elif wp.dtype != like.dtype:
+        wp = wp.to(like.dtype)
+    wp = wp.contiguous()
+    _cache_p[w] = (key, wp)
+    return wp
+
+
+def compute_weight_tp(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
+    """Padded channel-transposed weight ([I,kh,kw,O], dims >= 8)."""
+    ent = _cache_tp.get(w)
+    ver = w._version
+    key = (ver, like.dtype, "tp")
+    if ent is not None and ent[0] == key:
+        return ent[1]
+    wt = _pad_dims(w.detach()).permute(3, 1, 2, 0).contiguous()
+    if like.dtype == torch.bfloat16 and wt.dtype != torch.bfloat16:
+        wt = wt.to(torch.bfloat16)
+    elif wt.dtype != like.dtype:
+        wt = wt.to(like.dtype)
+    _cache_tp[w] = (key, wt)
+    return wt
 
 
 def compute_weight_t(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:

@@ -752,18 +752,35 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
     d_n[j] = n0 + cb / 2;
   }
 
+  // incremental m-decode (ow/oh/b advance by +WG_BM each m-step; no
+  // per-step division — the wrap loop runs <= WG_BM/OW + 1 iterations)
+  int a_ow[4], a_oh[4], a_b[4];
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    long m = mstart + a_row[j];
+    a_ow[j] = (int)(m % p.OW);
+    int t = (int)(m / p.OW);
+    a_oh[j] = t % p.OH;
+    a_b[j] = t / p.OH;
+  }
+
+  auto advance = [&]() {
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      a_ow[j] += WG_BM;
+      while (a_ow[j] >= p.OW) { a_ow[j] -= p.OW; ++a_oh[j]; }
+      while (a_oh[j] >= p.OH) { a_oh[j] -= p.OH; ++a_b[j]; }
+    }
+  };
+
   auto stage = [&](int buf, long ms) {
     #pragma unroll
     for (int j = 0; j < 4; ++j) {
       unsigned vo = 0xFFFFFF00u;
       long m = ms + a_row[j];
       if (a_kv[j] && m < p.M) {
-        int ow = (int)(m % p.OW);
-        int t = (int)(m / p.OW);
-        int oh = t % p.OH;
-        int b = t / p.OH;
-        int ih = oh * p.stride - p.pt + a_dkh[j];
-        int iw = ow * p.stride - p.pl + a_dkw[j];
+        int ih = a_oh[j] * p.stride - p.pt + a_dkh[j];
+        int iw = a_ow[j] * p.stride - p.pl + a_dkw[j];
         bool valid = true;
         if (p.reflect) {
           ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
@@ -771,7 +788,7 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
           valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
         }
         if (valid)
-          vo = (unsigned)(((((long)b * p.H + ih) * p.W + iw) * p.Cin + a_ci[j]) * 2);
+          vo = (unsigned)(((((long)a_b[j] * p.H + ih) * p.W + iw) * p.Cin + a_ci[j]) * 2);
       }
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
           rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * 4 + j) * 512],
@@ -803,7 +820,10 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
 
   int cur = 0;
   for (long ms = mstart; ms < mend; ms += WG_BM, cur ^= 1) {
-    if (ms + WG_BM < mend) stage(cur ^ 1, ms + WG_BM);
+    if (ms + WG_BM < mend) {
+      advance();
+      stage(cur ^ 1, ms + WG_BM);
+    }
     const char* Ab = (const char*)sm.A[cur];
     const char* Db = (const char*)sm.D[cur];
     #pragma unroll
@@ -859,23 +879,24 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
 
 // ---------------- reflect fold (dgrad border scatter) ----------------
 // dx[b,i,j,c] = Σ over padded positions q with mirror(q - pad) == (i,j)
+// vectorized: one thread folds 8 channels of one (b,i,j); the candidate
+// logic is per-pixel, shared by all 8 lanes' channels.
 __global__ void reflect_fold_kernel(const short* __restrict__ dxp,
-                                    float* __restrict__ dx_unused,
                                     short* __restrict__ dx,
                                     int B, int H, int W, int C,
                                     int pt, int pb, int pl, int pr) {
   long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long total = (long)B * H * W * C;
+  int gpr = C / 8;
+  long total = (long)B * H * W * gpr;
   if (idx >= total) return;
-  int c = (int)(idx % C);
-  long t = idx / C;
+  int g = (int)(idx % gpr);
+  long t = idx / gpr;
   int j = (int)(t % W);
   t /= W;
   int i = (int)(t % H);
   int b = (int)(t / H);
   int HP = H + pt + pb, WP = W + pl + pr;
-  float s = 0.f;
-  // candidate padded rows: identity, top mirror, bottom mirror
+  float s[8] = {};
   int hc[3] = {pt + i, pt - i, pt + 2 * (H - 1) - i};
   int wc_[3] = {pl + j, pl - j, pl + 2 * (W - 1) - j};
   #pragma unroll
@@ -892,10 +913,15 @@ __global__ void reflect_fold_kernel(const short* __restrict__ dxp,
       if (d > 0 && qw == wc_[0]) continue;
       if (d == 2 && qw == wc_[1]) continue;
       if (mirror_idx(qw - pl, W) != j) continue;
-      s += b2f(dxp[(((long)b * HP + qh) * WP + qw) * C + c]);
+      v8s v = *(const v8s*)(dxp + (((long)b * HP + qh) * WP + qw) * C + g * 8);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) s[e] += b2f(v[e]);
     }
   }
-  dx[idx] = f2b(s);
+  v8s out;
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) out[e] = f2b(s[e]);
+  *(v8s*)(dx + (((long)b * H + i) * W + j) * C + g * 8) = out;
 }
 
 // ================= host wrappers =================
@@ -1032,11 +1058,12 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wt, int64_t H, int64_t W,
   p.pt = 0; p.pl = 0;
   launch_conv(p, true, stream);
   auto dx = at::empty({dy.size(0), H, W, wt.size(0)}, dy.options());
-  long total = dx.numel();
+  TORCH_CHECK(wt.size(0) % 8 == 0, "reflect dgrad: Cin % 8 != 0");
+  long total = dx.numel() / 8;
   int threads = 256;
   hipLaunchKernelGGL(reflect_fold_kernel, dim3(cdiv(total, threads)),
                      dim3(threads), 0, stream,
-                     (const short*)dxp.const_data_ptr(), nullptr,
+                     (const short*)dxp.const_data_ptr(),
                      (short*)dx.mutable_data_ptr(),
                      (int)dy.size(0), (int)H, (int)W, (int)wt.size(0),
                      (int)pt, (int)pb, (int)pl, (int)pr);

@@ -89,7 +89,20 @@ __global__ __launch_bounds__(NT) void in_norm_kernel(
   __shared__ float sm[MAXC], sr[MAXC];
   for (int c = tid; c < C; c += NT) {
     float sv = 0, qv = 0;
-    for (int k = 0; k < S; ++k) {
+    int k = 0;
+    for (; k + 4 <= S; k += 4) {  // 4 independent chains hide L2 latency
+      float s0 = psum[((long)(k + 0) * B + b) * C + c];
+      float s1 = psum[((long)(k + 1) * B + b) * C + c];
+      float s2 = psum[((long)(k + 2) * B + b) * C + c];
+      float s3 = psum[((long)(k + 3) * B + b) * C + c];
+      float q0 = psq[((long)(k + 0) * B + b) * C + c];
+      float q1 = psq[((long)(k + 1) * B + b) * C + c];
+      float q2 = psq[((long)(k + 2) * B + b) * C + c];
+      float q3 = psq[((long)(k + 3) * B + b) * C + c];
+      sv += (s0 + s1) + (s2 + s3);
+      qv += (q0 + q1) + (q2 + q3);
+    }
+    for (; k < S; ++k) {
       sv += psum[((long)k * B + b) * C + c];
       qv += psq[((long)k * B + b) * C + c];
     }
@@ -201,7 +214,20 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
   __shared__ float sm1[MAXC], sm2[MAXC], smean[MAXC], srstd[MAXC];
   for (int c = tid; c < C; c += NT) {
     float t1 = 0, t2 = 0;
-    for (int k = 0; k < S; ++k) {
+    int k = 0;
+    for (; k + 4 <= S; k += 4) {
+      float a0 = p1[((long)(k + 0) * B + b) * C + c];
+      float a1 = p1[((long)(k + 1) * B + b) * C + c];
+      float a2 = p1[((long)(k + 2) * B + b) * C + c];
+      float a3 = p1[((long)(k + 3) * B + b) * C + c];
+      float b0 = p2[((long)(k + 0) * B + b) * C + c];
+      float b1 = p2[((long)(k + 1) * B + b) * C + c];
+      float b2 = p2[((long)(k + 2) * B + b) * C + c];
+      float b3 = p2[((long)(k + 3) * B + b) * C + c];
+      t1 += (a0 + a1) + (a2 + a3);
+      t2 += (b0 + b1) + (b2 + b3);
+    }
+    for (; k < S; ++k) {
       t1 += p1[((long)k * B + b) * C + c];
       t2 += p2[((long)k * B + b) * C + c];
     }
@@ -476,8 +502,9 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
   auto p1 = at::empty({S, B, C}, fopt);
   auto p2 = at::empty({S, B, C}, fopt);
   auto dx = at::empty_like(x);
-  auto dgamma = at::zeros({C}, fopt);
-  auto dbeta = at::zeros({C}, fopt);
+  auto dgb = at::zeros({2, C}, fopt);
+  auto dbeta = dgb[0];
+  auto dgamma = dgb[1];
   hipLaunchKernelGGL(in_bwd_reduce_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)dy.const_data_ptr(),
                      (const short*)x.const_data_ptr(),

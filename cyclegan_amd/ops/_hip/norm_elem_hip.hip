@@ -44,24 +44,24 @@ __global__ __launch_bounds__(NT) void in_reduce_kernel(
       s[j] += f; q[j] += f * f;
     }
   }
-  // reduce across the rstep threads sharing this channel group via LDS
-  __shared__ float red[NT * 2];
-  long out = ((long)sl * B + b) * C + g * 8;
+  // one-barrier reduce: every thread parks its 16 partials in LDS, then
+  // C*2 output elements are summed in parallel.
+  __shared__ float red[NT * 17];  // +1 pad: conflict-free strided reads
   #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    red[tid] = s[j];
-    red[NT + tid] = q[j];
-    __syncthreads();
-    if (rof == 0) {
-      float ts = 0, tq = 0;
-      for (int k = 0; k < rstep; ++k) {
-        ts += red[g + k * gpr];
-        tq += red[NT + g + k * gpr];
-      }
-      psum[out + j] = ts;
-      psq[out + j] = tq;
-    }
-    __syncthreads();
+    red[tid * 17 + j] = s[j];
+    red[tid * 17 + 8 + j] = q[j];
+  }
+  __syncthreads();
+  long out = ((long)sl * B + b) * C;
+  for (int o = tid; o < C * 2; o += NT) {
+    int c = o % C;
+    int issq = o / C;
+    int g2 = c / 8, j = c % 8;
+    float t = 0;
+    for (int k = 0; k < rstep; ++k)
+      t += red[(g2 + k * gpr) * 17 + issq * 8 + j];
+    (issq ? psq : psum)[out + c] = t;
   }
 }
 
@@ -119,22 +119,26 @@ __global__ __launch_bounds__(NT) void in_norm_kernel(
   }
   __syncthreads();
 
-  for (long e = tid; e < (r1 - r0) * gpr; e += NT) {
-    long r = r0 + e / gpr;
-    int g = (int)(e % gpr);
-    long off = base + r * C + g * 8;
-    v8s v = *(const v8s*)(x + off);
-    v8s rv = {};
-    if (res) rv = *(const v8s*)(res + off);
-    v8s out;
-    #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int c = g * 8 + j;
-      float val = (b2f(v[j]) - sm[c]) * sr[c] * gamma[c] + beta[c];
-      if (res) val += b2f(rv[j]);
-      out[j] = f2b(apply_act(val, act, slope));
+  {
+    int g = tid % gpr;
+    int rstep = NT / gpr;
+    int rof = tid / gpr;
+    #pragma unroll 1
+    for (long r = r0 + rof; r < r1; r += rstep) {
+      long off = base + r * C + g * 8;
+      v8s v = *(const v8s*)(x + off);
+      v8s rv = {};
+      if (res) rv = *(const v8s*)(res + off);
+      v8s out;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int c = g * 8 + j;
+        float val = (b2f(v[j]) - sm[c]) * sr[c] * gamma[c] + beta[c];
+        if (res) val += b2f(rv[j]);
+        out[j] = f2b(apply_act(val, act, slope));
+      }
+      *(v8s*)(y + off) = out;
     }
-    *(v8s*)(y + off) = out;
   }
 }
 
@@ -173,23 +177,22 @@ __global__ __launch_bounds__(NT) void in_bwd_reduce_kernel(
       a1[j] += d; a2[j] += d * xh;
     }
   }
-  __shared__ float red[NT * 2];
-  long out = ((long)sl * B + b) * C + g * 8;
+  __shared__ float red[NT * 17];
   #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    red[tid] = a1[j];
-    red[NT + tid] = a2[j];
-    __syncthreads();
-    if (rof == 0) {
-      float t1 = 0, t2 = 0;
-      for (int k = 0; k < rstep; ++k) {
-        t1 += red[g + k * gpr];
-        t2 += red[NT + g + k * gpr];
-      }
-      p1[out + j] = t1;
-      p2[out + j] = t2;
-    }
-    __syncthreads();
+    red[tid * 17 + j] = a1[j];
+    red[tid * 17 + 8 + j] = a2[j];
+  }
+  __syncthreads();
+  long out = ((long)sl * B + b) * C;
+  for (int o = tid; o < C * 2; o += NT) {
+    int c = o % C;
+    int is2 = o / C;
+    int g2 = c / 8, j = c % 8;
+    float t = 0;
+    for (int k = 0; k < rstep; ++k)
+      t += red[(g2 + k * gpr) * 17 + is2 * 8 + j];
+    (is2 ? p2 : p1)[out + c] = t;
   }
 }
 
@@ -238,22 +241,26 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
   }
   __syncthreads();
 
-  for (long e = tid; e < (r1 - r0) * gpr; e += NT) {
-    long r = r0 + e / gpr;
-    int g = (int)(e % gpr);
-    long off = base + r * C + g * 8;
-    v8s dv = *(const v8s*)(dy + off);
-    v8s xv = *(const v8s*)(x + off);
-    v8s out;
-    #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int c = g * 8 + j;
-      float rs = srstd[c];
-      float xh = (b2f(xv[j]) - smean[c]) * rs;
-      float d = b2f(dv[j]);
-      out[j] = f2b(gamma[c] * rs * (d - sm1[c] - xh * sm2[c]));
+  {
+    int g = tid % gpr;
+    int rstep = NT / gpr;
+    int rof = tid / gpr;
+    #pragma unroll 1
+    for (long r = r0 + rof; r < r1; r += rstep) {
+      long off = base + r * C + g * 8;
+      v8s dv = *(const v8s*)(dy + off);
+      v8s xv = *(const v8s*)(x + off);
+      v8s out;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int c = g * 8 + j;
+        float rs = srstd[c];
+        float xh = (b2f(xv[j]) - smean[c]) * rs;
+        float d = b2f(dv[j]);
+        out[j] = f2b(gamma[c] * rs * (d - sm1[c] - xh * sm2[c]));
+      }
+      *(v8s*)(dx + off) = out;
     }
-    *(v8s*)(dx + off) = out;
   }
 }
 

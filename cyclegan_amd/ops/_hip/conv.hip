@@ -402,6 +402,7 @@ struct WgradParams {
   const short* __restrict__ x;    // [B,H,W,Cin]
   const short* __restrict__ dy;   // [B,OH,OW,Cout]
   float* __restrict__ dw;         // [Cout,KH,KW,Cin] fp32 (pre-zeroed)
+  float* __restrict__ ws;         // [slices*ktiles*ntiles][WG_BN][WG_BK] slabs
   int B, H, W, Cin, OH, OW, Cout, KH, KW;
   int stride, pt, pl, reflect;
   long M, KTOT;
@@ -861,20 +862,46 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
     __syncthreads();
   }
 
-  #pragma unroll
-  for (int nf = 0; nf < 2; ++nf) {
-    int n = n0 + wn0 + nf * 16 + fr;
-    if (n >= p.Cout) continue;
+  // slab store (no atomics): chunk layout [WG_BN][WG_BK], float4 rows
+  {
+    long chunk = (((long)sl * p.ktiles + kt) * p.ntiles + nt) *
+                 ((long)WG_BN * WG_BK);
     #pragma unroll
-    for (int kf = 0; kf < 4; ++kf) {
+    for (int nf = 0; nf < 2; ++nf) {
+      int nl = wn0 + nf * 16 + fr;
       #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        long k = k0 + wk0 + kf * 16 + fg * 4 + r;
-        if (k < p.KTOT)
-          atomicAdd(&p.dw[(long)n * p.KTOT + k], acc[kf][nf][r]);
+      for (int kf = 0; kf < 4; ++kf) {
+        int kl = wk0 + kf * 16 + fg * 4;
+        *(float4*)&p.ws[chunk + (long)nl * WG_BK + kl] =
+            *(const float4*)&acc[kf][nf];
       }
     }
   }
+}
+
+// sum the split-M slabs into dw (coalesced; also applies the K/N guards)
+__global__ void wgrad_reduce_kernel(const float* __restrict__ ws,
+                                    float* __restrict__ dw, long KTOT,
+                                    int Cout, int ktiles, int ntiles,
+                                    int slices) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long total = (long)ktiles * WG_BK * ntiles * WG_BN;
+  if (idx >= total) return;
+  // idx -> (nt, nl, kt, kl) with kl fastest for coalescing
+  int kl = (int)(idx % WG_BK);
+  long t = idx / WG_BK;
+  int kt = (int)(t % ktiles);
+  t /= ktiles;
+  int nl = (int)(t % WG_BN);
+  int nt = (int)(t / WG_BN);
+  long k = (long)kt * WG_BK + kl;
+  int n = nt * WG_BN + nl;
+  if (k >= KTOT || n >= Cout) return;
+  long stride = (long)ktiles * ntiles * WG_BN * WG_BK;
+  long off = (((long)kt * ntiles + nt) * WG_BN + nl) * WG_BK + kl;
+  float s = 0;
+  for (int sl = 0; sl < slices; ++sl) s += ws[off + sl * stride];
+  dw[(long)n * KTOT + k] = s;
 }
 
 // ---------------- reflect fold (dgrad border scatter) ----------------
@@ -1098,25 +1125,41 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
   p.KTOT = (long)KH * KW * p.Cin;
   p.ktiles = cdiv(p.KTOT, WG_BK);
   p.ntiles = cdiv(p.Cout, WG_BN);
-  // split M so total blocks ≈ 2-4x CU count
+  // split M so total blocks ≈ 2-4x CU count (CYG_WG_BLOCKS to sweep)
   long mchunks = (p.M + WG_BM - 1) / WG_BM;
-  int target = std::max<long>(1, 512 / ((long)p.ktiles * p.ntiles));
+  static int tgt_blocks = []() {
+    const char* e = getenv("CYG_WG_BLOCKS");
+    return e ? atoi(e) : 512;
+  }();
+  int target = std::max<long>(1, tgt_blocks / ((long)p.ktiles * p.ntiles));
   int slices = (int)std::min<long>(mchunks, target);
   p.mchunks_per_slice = (mchunks + slices - 1) / slices;
   p.slices = slices;
-  auto dw = at::zeros({p.Cout, (long)KH, (long)KW, p.Cin},
-                      x.options().dtype(at::kFloat));
-  p.dw = (float*)dw.mutable_data_ptr();
   dim3 grid((long)p.ktiles * p.ntiles * p.slices);
   bool glds_ok = (p.Cin % 8) == 0 && (p.Cout % 8) == 0 &&
                  (long)p.B * p.H * p.W * p.Cin * 2 < (1L << 31) &&
                  p.M * p.Cout * 2 < (1L << 31);
-  if (glds_ok)
-    hipLaunchKernelGGL(wgrad_glds_kernel, grid, dim3(NTHREADS), 0,
-                       at::cuda::getCurrentCUDAStream(), p);
-  else
-    hipLaunchKernelGGL(wgrad_kernel, grid, dim3(NTHREADS), 0,
-                       at::cuda::getCurrentCUDAStream(), p);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (glds_ok) {
+    auto dw = at::empty({p.Cout, (long)KH, (long)KW, p.Cin},
+                        x.options().dtype(at::kFloat));
+    p.dw = (float*)dw.mutable_data_ptr();
+    auto ws = at::empty({(long)p.slices * p.ktiles * p.ntiles *
+                         WG_BN * WG_BK},
+                        x.options().dtype(at::kFloat));
+    p.ws = (float*)ws.mutable_data_ptr();
+    hipLaunchKernelGGL(wgrad_glds_kernel, grid, dim3(NTHREADS), 0, stream, p);
+    long total = (long)p.ktiles * WG_BK * p.ntiles * WG_BN;
+    hipLaunchKernelGGL(wgrad_reduce_kernel, dim3(cdiv(total, 256)),
+                       dim3(256), 0, stream,
+                       (const float*)ws.const_data_ptr(), p.dw, p.KTOT,
+                       p.Cout, p.ktiles, p.ntiles, p.slices);
+    return dw;
+  }
+  auto dw = at::zeros({p.Cout, (long)KH, (long)KW, p.Cin},
+                      x.options().dtype(at::kFloat));
+  p.dw = (float*)dw.mutable_data_ptr();
+  hipLaunchKernelGGL(wgrad_kernel, grid, dim3(NTHREADS), 0, stream, p);
   return dw;
 }
 

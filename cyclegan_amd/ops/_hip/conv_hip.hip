@@ -296,45 +296,97 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
       (void*)p.w, 0, (unsigned)(p.Cout * p.KTOT * 2), 0x00020000);
 
   const int nk = (int)((p.KTOT + BK - 1) / BK);
+  const bool big_ci = p.Cin >= BK;  // wave-uniform
 
-  auto stage = [&](int buf, long k0) {
-    long k = k0 + klog;
-    bool kv = k < p.KTOT;
-    int tap = kv ? (int)(k / p.Cin) : 0;
-    int ci = (int)(k - (long)tap * p.Cin);
-    int dkh = tap / p.KW, dkw = tap - (tap / p.KW) * p.KW;
+  // incremental gather state: recompute voffsets only at tap boundaries
+  // (every Cin/64 K-steps); otherwise one predicated add per instruction.
+  long kcur;
+  bool kv;
+  int ci, dkh, dkw;
+  unsigned avo[4];
+  bool avalid[4];
+  unsigned bvo[2];
+  bool bnv[2];
+
+  auto recompute_a = [&]() {
     #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      unsigned vo = 0xFFFFFF00u;
-      if (kv && aok[j]) {
-        bool valid = true;
-        int ih, iw;
-        if (IS_CONVT) {
-          int nh = aih[j] - dkh, nw = aiw[j] - dkw;
-          valid = nh >= 0 && nw >= 0 && (nh % stride) == 0 &&
-                  (nw % stride) == 0;
-          ih = nh / stride; iw = nw / stride;
-          valid = valid && ih < p.H && iw < p.W;
+      avalid[j] = false;
+      avo[j] = 0xFF000000u;
+      if (!aok[j]) continue;
+      bool valid = true;
+      int ih, iw;
+      if (IS_CONVT) {
+        int nh = aih[j] - dkh, nw = aiw[j] - dkw;
+        valid = nh >= 0 && nw >= 0 && (nh % stride) == 0 && (nw % stride) == 0;
+        ih = nh / stride; iw = nw / stride;
+        valid = valid && ih < p.H && iw < p.W;
+      } else {
+        ih = aih[j] + dkh; iw = aiw[j] + dkw;
+        if (p.reflect) {
+          ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
         } else {
-          ih = aih[j] + dkh; iw = aiw[j] + dkw;
-          if (p.reflect) {
-            ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
-          } else {
-            valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
-          }
+          valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
         }
-        if (valid)
-          vo = axb[j] + (unsigned)((((long)ih * p.W + iw) * p.Cin + ci) * 2);
       }
+      if (valid) {
+        avalid[j] = true;
+        avo[j] = axb[j] + (unsigned)((((long)ih * p.W + iw) * p.Cin + ci) * 2);
+      }
+    }
+  };
+
+  auto init_state = [&]() {
+    kcur = klog;
+    kv = kcur < p.KTOT;
+    int tap = (int)(kcur / p.Cin);
+    ci = (int)(kcur - (long)tap * p.Cin);
+    dkh = tap / p.KW;
+    dkw = tap - dkh * p.KW;
+    recompute_a();
+    #pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      bnv[j] = bn[j] < p.Cout;
+      bvo[j] = (unsigned)(((long)bn[j] * p.KTOT + kcur) * 2);
+    }
+  };
+
+  auto advance = [&]() {
+    kcur += BK;
+    kv = kcur < p.KTOT;
+    #pragma unroll
+    for (int j = 0; j < 2; ++j) bvo[j] += BK * 2;
+    if (big_ci) {
+      ci += BK;
+      if (ci >= p.Cin) {
+        ci -= p.Cin;
+        if (++dkw == p.KW) { dkw = 0; ++dkh; }
+        recompute_a();
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) avo[j] += BK * 2;
+      }
+    } else {
+      // Cin < 64: a K-step crosses several taps — full recompute
+      int tap = (int)(kcur / p.Cin);
+      ci = (int)(kcur - (long)tap * p.Cin);
+      dkh = tap / p.KW;
+      dkw = tap - dkh * p.KW;
+      recompute_a();
+    }
+  };
+
+  auto stage = [&](int buf) {
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      unsigned vo = (kv && avalid[j]) ? avo[j] : 0xFF000000u;
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
           rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * 4 + j) * 512],
           16, vo, 0, 0, 0);
     }
     #pragma unroll
     for (int j = 0; j < 2; ++j) {
-      unsigned vo = 0xFFFFFF00u;
-      if (k < p.KTOT && bn[j] < p.Cout)
-        vo = (unsigned)(((long)bn[j] * p.KTOT + k) * 2);
+      unsigned vo = (kv && bnv[j]) ? bvo[j] : 0xFF000000u;
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
           rw, (__attribute__((address_space(3))) void*)&sm.Bt[buf][(w * 2 + j) * 512],
           16, vo, 0, 0, 0);
@@ -348,11 +400,15 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
   const int fg = lane >> 4;
   const int swz = (fr & 7) << 4;  // read-side XOR (bytes)
 
-  stage(0, 0);
+  init_state();
+  stage(0);
   __syncthreads();
 
   for (int kt = 0; kt < nk; ++kt) {
-    if (kt + 1 < nk) stage((kt + 1) & 1, (long)(kt + 1) * BK);
+    if (kt + 1 < nk) {
+      advance();
+      stage((kt + 1) & 1);
+    }
     const char* Ab = (const char*)sm.A[kt & 1];
     const char* Bb = (const char*)sm.Bt[kt & 1];
     #pragma unroll

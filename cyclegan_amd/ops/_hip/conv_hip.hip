@@ -450,6 +450,214 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
   }
 }
 
+// ---------------- phase-decomposed stride-2 transpose conv ----------------
+// convT s2 (and s2 conv dgrad) gathers hit only taps whose parity matches
+// the output pixel's phase — 3/4 of a dense gather is zeros. Here the M
+// space is regrouped into 4 phase classes (p.mtiles = 4 * tiles/phase, OH
+// and OW even), each tile enumerates ONLY its phase's taps
+// (K_eff = nvh*nvw*Cin), so every MFMA operates on useful data.
+__global__ __launch_bounds__(NTHREADS) void convt_phased_kernel(ConvParams p) {
+  __shared__ ConvSmem sm;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int bid = blockIdx.x;
+  const int mt = bid % p.mtiles, nt = bid / p.mtiles;
+  const int tpp = p.mtiles >> 2;          // tiles per phase
+  const int phase = mt / tpp;
+  const int ph = phase >> 1, pw = phase & 1;
+  const int OH2 = p.OH >> 1, OW2 = p.OW >> 1;
+  const long Mp = (long)p.B * OH2 * OW2;  // pixels per phase
+  const long m0 = (long)(mt - phase * tpp) * BM;
+  const int n0 = nt * BN;
+  // valid tap counts for this phase: dk = 2*v + parity, parity = (ph+pt)&1
+  const int par_h = (ph + p.pt) & 1 ? ((p.pt + 1) & 1) : 0;  // see below
+  // dkh must satisfy (i + pt - dkh) even where (i+pt)%2 == phh_sel;
+  // phases are indexed directly by (i+pt)&1, so dkh parity == phase bit.
+  const int dkh_par = ph, dkw_par = pw;
+  const int nvh = (p.KH - dkh_par + 1) >> 1;
+  const int nvw = (p.KW - dkw_par + 1) >> 1;
+  const long KEFF = (long)nvh * nvw * p.Cin;
+  (void)par_h;
+
+  for (int r = tid; r < BM; r += NTHREADS) {
+    long mp = m0 + r;
+    bool ok = mp < Mp;
+    long mm = ok ? mp : 0;
+    int owp = (int)(mm % OW2);
+    int ohp = (int)((mm / OW2) % OH2);
+    int b = (int)(mm / ((long)OW2 * OH2));
+    // output coords of this phase: (i+pt)&1 == ph, (j+pl)&1 == pw
+    int i = 2 * ohp + ((ph + p.pt) & 1);
+    int j = 2 * owp + ((pw + p.pl) & 1);
+    sm.rowok[r] = ok && i < p.OH && j < p.OW;
+    sm.rowxb[r] = (unsigned)((long)b * p.H * p.W * p.Cin * 2);
+    sm.rowyb[r] = ((long)(b * p.OH + i) * p.OW + j) * p.Cout;
+    sm.rowih[r] = (i + p.pt - dkh_par) >> 1;  // oh = this - vh
+    sm.rowiw[r] = (j + p.pl - dkw_par) >> 1;
+  }
+  __syncthreads();
+
+  const int lr = lane >> 3;
+  const int klog = ((lane & 7) ^ lr) << 3;
+  int aih[4], aiw[4];
+  unsigned axb[4];
+  bool aok[4];
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int r = w * 32 + j * 8 + lr;
+    aih[j] = sm.rowih[r];
+    aiw[j] = sm.rowiw[r];
+    axb[j] = sm.rowxb[r];
+    aok[j] = sm.rowok[r];
+  }
+  const int bn[2] = {(w * 2 + 0) * 8 + lr + n0, (w * 2 + 1) * 8 + lr + n0};
+
+  auto rx = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)p.x, 0, (unsigned)((long)p.B * p.H * p.W * p.Cin * 2), 0x00020000);
+  auto rw = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)p.w, 0, (unsigned)(p.Cout * p.KTOT * 2), 0x00020000);
+
+  const int nk = (int)((KEFF + BK - 1) / BK);
+  const bool big_ci = p.Cin >= BK;
+
+  long kcur;
+  bool kv;
+  int ci, vh, vw;
+  unsigned avo[4], bvo[2];
+  bool avalid[4], bnv[2];
+
+  auto recompute = [&]() {
+    int oh_off = vh, ow_off = vw;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      avalid[j] = false;
+      avo[j] = 0xFF000000u;
+      if (!aok[j]) continue;
+      int oh = aih[j] - oh_off, ow = aiw[j] - ow_off;
+      if (oh >= 0 && oh < p.H && ow >= 0 && ow < p.W) {
+        avalid[j] = true;
+        avo[j] = axb[j] + (unsigned)((((long)oh * p.W + ow) * p.Cin + ci) * 2);
+      }
+    }
+    // weight k index for this tap
+    long kw_idx = ((long)(2 * vh + dkh_par) * p.KW + (2 * vw + dkw_par)) * p.Cin + ci;
+    #pragma unroll
+    for (int j = 0; j < 2; ++j)
+      bvo[j] = (unsigned)(((long)bn[j] * p.KTOT + kw_idx) * 2);
+  };
+
+  auto init_state = [&]() {
+    kcur = klog;
+    kv = kcur < KEFF;
+    int vtap = (int)(kcur / p.Cin);
+    ci = (int)(kcur - (long)vtap * p.Cin);
+    vh = vtap / nvw;
+    vw = vtap - vh * nvw;
+    #pragma unroll
+    for (int j = 0; j < 2; ++j) bnv[j] = bn[j] < p.Cout;
+    recompute();
+  };
+
+  auto advance = [&]() {
+    kcur += BK;
+    kv = kcur < KEFF;
+    if (big_ci) {
+      ci += BK;
+      if (ci >= p.Cin) {
+        ci -= p.Cin;
+        if (++vw == nvw) { vw = 0; ++vh; }
+        recompute();
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) avo[j] += BK * 2;
+        #pragma unroll
+        for (int j = 0; j < 2; ++j) bvo[j] += BK * 2;
+      }
+    } else {
+      int vtap = (int)(kcur / p.Cin);
+      ci = (int)(kcur - (long)vtap * p.Cin);
+      vh = vtap / nvw;
+      vw = vtap - vh * nvw;
+      recompute();
+    }
+  };
+
+  auto stage = [&](int buf) {
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      unsigned vo = (kv && avalid[j]) ? avo[j] : 0xFF000000u;
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * 4 + j) * 512],
+          16, vo, 0, 0, 0);
+    }
+    #pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      unsigned vo = (kv && bnv[j]) ? bvo[j] : 0xFF000000u;
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rw, (__attribute__((address_space(3))) void*)&sm.Bt[buf][(w * 2 + j) * 512],
+          16, vo, 0, 0, 0);
+    }
+  };
+
+  v4f acc[4][2] = {};
+  const int wr = w >> 1, wc = w & 1;
+  const int wm0 = wr * 64, wn0 = wc * 32;
+  const int fr = lane & 15;
+  const int fg = lane >> 4;
+  const int swz = (fr & 7) << 4;
+
+  init_state();
+  stage(0);
+  __syncthreads();
+
+  for (int kt = 0; kt < nk; ++kt) {
+    if (kt + 1 < nk) {
+      advance();
+      stage((kt + 1) & 1);
+    }
+    const char* Ab = (const char*)sm.A[kt & 1];
+    const char* Bb = (const char*)sm.Bt[kt & 1];
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 32) {
+      const int kbyte = (kk + fg * 8) * 2;
+      v8bf a0 = *(const v8bf*)(Ab + ((wm0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
+      v8bf a1 = *(const v8bf*)(Ab + ((wm0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
+      v8bf a2 = *(const v8bf*)(Ab + ((wm0 + 2 * 16 + fr) << 7) + (kbyte ^ swz));
+      v8bf a3 = *(const v8bf*)(Ab + ((wm0 + 3 * 16 + fr) << 7) + (kbyte ^ swz));
+      v8bf b0 = *(const v8bf*)(Bb + ((wn0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
+      v8bf b1 = *(const v8bf*)(Bb + ((wn0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
+      acc[2][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b0, acc[2][0], 0, 0, 0);
+      acc[2][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b1, acc[2][1], 0, 0, 0);
+      acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b0, acc[3][0], 0, 0, 0);
+      acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b1, acc[3][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  #pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+    int n = n0 + wn0 + nf * 16 + fr;
+    if (n >= p.Cout) continue;
+    float bv = p.bias ? b2f(p.bias[n]) : 0.f;
+    #pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int rl = wm0 + mf * 16 + fg * 4 + r;
+        if (!sm.rowok[rl]) continue;
+        float v = apply_act(acc[mf][nf][r] + bv, p.act, p.slope);
+        p.y[sm.rowyb[rl] + n] = f2b(v);
+      }
+    }
+  }
+}
+
 // ---------------- weight gradient ----------------
 // dw[n][k] = Σ_m A[m][k] · dy[m][n];  A-tile and dy-tile staged TRANSPOSED
 // (m contiguous per row) so the MFMA reduce dim is m. fp32 atomics over
@@ -1053,6 +1261,16 @@ static void launch_conv(const ConvParams& p, bool is_convt, hipStream_t stream) 
   bool glds_ok = aligned &&
                  (long)p.B * p.H * p.W * p.Cin * 2 < (1L << 31) &&
                  (long)p.Cout * p.KTOT * 2 < (1L << 31);
+  if (glds_ok && is_convt && p.stride == 2 && (p.OH % 2) == 0 &&
+      (p.OW % 2) == 0) {
+    // phase-decomposed: 4 phase classes, only matching-parity taps
+    ConvParams q = p;
+    long Mp = (long)q.B * (q.OH / 2) * (q.OW / 2);
+    q.mtiles = 4 * cdiv(Mp, BM);
+    dim3 g2((long)q.mtiles * q.ntiles);
+    hipLaunchKernelGGL(convt_phased_kernel, g2, dim3(NTHREADS), 0, stream, q);
+    return;
+  }
   if (glds_ok) {
     if (is_convt) launch_glds_s<true>(p, grid, stream);
     else launch_glds_s<false>(p, grid, stream);

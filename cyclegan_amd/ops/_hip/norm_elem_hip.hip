@@ -452,7 +452,8 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
 // ================= host wrappers =================
 
 static int slices_for(long HW, int B) {
-  // aim for >= 512 blocks overall
+  // >= ~512 blocks for occupancy, but cap slices: slab traffic in the
+  // stats phases scales with S per block.
   int s = (int)std::min<long>(std::max<long>(1, 512 / std::max(B, 1)),
                               std::max<long>(1, HW / 64));
   return std::max(1, s);

@@ -93,8 +93,14 @@ class CycleGAN:
         for g in self.groups.values():
             g.zero_grad()
 
-        fake_y = self.G(x)
-        fake_x = self.F(y)
+        b = x.shape[0]
+        # batched generator calls: every op is per-sample (convs, per-sample
+        # InstanceNorm stats, per-sample losses), so G(cat(x,y)) is
+        # numerically identical to G(x), G(y) — fewer, larger kernels.
+        g_out = self.G(torch.cat([x, y]))     # -> fake_y, same_y
+        fake_y, same_y = g_out[:b], g_out[b:]
+        f_out = self.F(torch.cat([y, x]))     # -> fake_x, same_x
+        fake_x, same_x = f_out[:b], f_out[b:]
 
         discriminate_fake_x = self.X(fake_x)
         discriminate_fake_y = self.Y(fake_y)
@@ -103,14 +109,17 @@ class CycleGAN:
         F_loss = self.generator_loss(discriminate_fake_x)
         G_cycle_loss = self.cycle_loss(y, self.G(fake_x))
         F_cycle_loss = self.cycle_loss(x, self.F(fake_y))
-        G_identity_loss = self.identity_loss(y, self.G(y))
-        F_identity_loss = self.identity_loss(x, self.F(x))
+        G_identity_loss = self.identity_loss(y, same_y)
+        F_identity_loss = self.identity_loss(x, same_x)
         G_total = G_loss + G_cycle_loss + G_identity_loss
         F_total = F_loss + F_cycle_loss + F_identity_loss
 
-        # discriminator pass on real + re-discriminated (detached) fakes
-        X_loss = self.discriminator_loss(self.X(x), self.X(fake_x.detach()))
-        Y_loss = self.discriminator_loss(self.Y(y), self.Y(fake_y.detach()))
+        # discriminator pass on real + re-discriminated (detached) fakes,
+        # batched (again identical per-sample math)
+        dx = self.X(torch.cat([x, fake_x.detach()]))
+        dy_ = self.Y(torch.cat([y, fake_y.detach()]))
+        X_loss = self.discriminator_loss(dx[:b], dx[b:])
+        Y_loss = self.discriminator_loss(dy_[:b], dy_[b:])
 
         torch.autograd.backward(G_total, inputs=self.groups["G"].params,
                                 retain_graph=True)

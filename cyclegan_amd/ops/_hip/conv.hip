@@ -231,19 +231,21 @@ __global__ __launch_bounds__(NTHREADS) void conv_gemm_kernel(ConvParams p) {
 // the buffer descriptor turns into zeros. One __shared__ object only
 // (hipcc de-pipelines glds next to a second one).
 
-struct ConvSmem {
-  short A[2][BM * BK];
-  short Bt[2][BN * BK];
+template <int NBUF>
+struct ConvSmemT {
+  short A[NBUF][BM * BK];
+  short Bt[NBUF][BN * BK];
   long rowyb[BM];
   int rowih[BM], rowiw[BM];
   unsigned rowxb[BM];  // byte offset of batch base (tensors < 4 GiB)
   char rowok[BM];
 };
+using ConvSmem = ConvSmemT<2>;
 
-template <bool IS_CONVT, int STRIDE>
+template <bool IS_CONVT, int STRIDE, int NBUF = 2>
 __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
   const int stride = STRIDE ? STRIDE : p.stride;
-  __shared__ ConvSmem sm;
+  __shared__ ConvSmemT<NBUF> sm;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -456,6 +458,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
 // and OW even), each tile enumerates ONLY its phase's taps
 // (K_eff = nvh*nvw*Cin), so every MFMA operates on useful data.
 __global__ __launch_bounds__(NTHREADS) void convt_phased_kernel(ConvParams p) {
+  constexpr int NBUF = 2;  // (shares the templated loop body text)
   __shared__ ConvSmem sm;
 
   const int tid = threadIdx.x;
@@ -608,16 +611,35 @@ __global__ __launch_bounds__(NTHREADS) void convt_phased_kernel(ConvParams p) {
   const int swz = (fr & 7) << 4;
 
   init_state();
-  stage(0);
-  __syncthreads();
+  if constexpr (NBUF == 2) {
+    stage(0);
+    __syncthreads();
+  } else {
+    // 3-buffer: keep one tile's LDS-DMA in flight ACROSS the barrier
+    // (counted vmcnt + raw s_barrier; __syncthreads would drain it)
+    stage(0);
+    if (nk > 1) { advance(); stage(1); }
+  }
 
   for (int kt = 0; kt < nk; ++kt) {
-    if (kt + 1 < nk) {
-      advance();
-      stage((kt + 1) & 1);
+    if constexpr (NBUF == 3) {
+      if (kt + 1 < nk)
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");  // tile kt landed
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      if (kt + 2 < nk) {
+        advance();
+        stage((kt + 2) % 3);
+      }
+    } else {
+      if (kt + 1 < nk) {
+        advance();
+        stage((kt + 1) & 1);
+      }
     }
-    const char* Ab = (const char*)sm.A[kt & 1];
-    const char* Bb = (const char*)sm.Bt[kt & 1];
+    const char* Ab = (const char*)sm.A[NBUF == 3 ? kt % 3 : (kt & 1)];
+    const char* Bb = (const char*)sm.Bt[NBUF == 3 ? kt % 3 : (kt & 1)];
     #pragma unroll
     for (int kk = 0; kk < BK; kk += 32) {
       const int kbyte = (kk + fg * 8) * 2;
@@ -636,8 +658,9 @@ __global__ __launch_bounds__(NTHREADS) void convt_phased_kernel(ConvParams p) {
       acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b0, acc[3][0], 0, 0, 0);
       acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b1, acc[3][1], 0, 0, 0);
     }
-    __syncthreads();
+    if constexpr (NBUF == 2) __syncthreads();
   }
+  if constexpr (NBUF == 3) __syncthreads();  // before epilogue rowyb reads
 
   #pragma unroll
   for (int nf = 0; nf < 2; ++nf) {
@@ -1236,8 +1259,32 @@ static void launch_conv_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
   }
 }
 
+static bool use_pipe3() {
+  static int v = []() {
+    const char* e = getenv("CYG_CONV_PIPE3");
+    return e ? atoi(e) : 0;
+  }();
+  return v != 0;
+}
+
 template <bool IS_CONVT>
 static void launch_glds_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
+  if (use_pipe3()) {
+    switch (p.stride) {
+      case 1:
+        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 1, 3>), grid,
+                           dim3(NTHREADS), 0, stream, p);
+        return;
+      case 2:
+        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 2, 3>), grid,
+                           dim3(NTHREADS), 0, stream, p);
+        return;
+      default:
+        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 0, 3>), grid,
+                           dim3(NTHREADS), 0, stream, p);
+        return;
+    }
+  }
   switch (p.stride) {
     case 1:
       hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 1>), grid,

@@ -24,14 +24,14 @@ from tqdm import tqdm
 
 from cyclegan_amd.parallel import DistContext
 from cyclegan_amd.trainer import CycleGAN
-from cyclegan_amd.data import Pipeline
+from cyclegan_amd.data import Pipeline, DevicePrefetcher
 from cyclegan_amd import utils
 
 
 def train(args, pipe, gan, summary, epoch: int):
     results = {}
-    for x, y in tqdm(pipe.train_epoch(epoch), desc="Train",
-                     total=pipe.train_steps,
+    it = DevicePrefetcher(pipe.train_epoch(epoch), gan.device, gan.compute_dtype)
+    for x, y in tqdm(it, desc="Train", total=pipe.train_steps,
                      disable=args.verbose == 0 or not gan.ctx.is_main):
         result = gan.train_step(x, y)
         utils.append_dict(results, result)
@@ -43,7 +43,8 @@ def train(args, pipe, gan, summary, epoch: int):
 
 def test(args, pipe, gan, summary, epoch: int):
     results = {}
-    for x, y in tqdm(pipe.test_epoch(), desc="Test", total=pipe.test_steps,
+    it = DevicePrefetcher(pipe.test_epoch(), gan.device, gan.compute_dtype)
+    for x, y in tqdm(it, desc="Test", total=pipe.test_steps,
                      disable=args.verbose == 0 or not gan.ctx.is_main):
         result = gan.test_step(x, y)
         utils.append_dict(results, result)

@@ -78,3 +78,14 @@ def test_plot_pairs(local_ctx):
     pairs = list(p.plot_pairs())
     assert len(pairs) == 4  # min(5, num_test)
     assert pairs[0][0].shape == (1, 64, 64, 3)
+
+
+def test_device_prefetcher_cpu_passthrough(local_ctx):
+    import torch
+    from cyclegan_amd.data import DevicePrefetcher
+    p = Pipeline(make_args(), local_ctx, image_size=64)
+    it = DevicePrefetcher(p.train_epoch(0), torch.device("cpu"), torch.float32)
+    batches = list(it)
+    assert len(batches) == p.train_steps
+    ref = list(p.train_epoch(0))
+    assert torch.equal(batches[0][0], ref[0][0])

@@ -34,7 +34,7 @@ def main():
     ap.add_argument("--batch_size", type=int, default=4, help="per-GPU batch")
     ap.add_argument("--image_size", type=int, default=256)
     ap.add_argument("--num_residual_blocks", type=int, default=9)
-    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32", "fp8"])
     ap.add_argument("--output_dir", default="/tmp/cyclegan_bench")
     args = ap.parse_args()
 
@@ -50,7 +50,9 @@ def main():
 
     torch.manual_seed(1234)
     args.global_batch_size = ctx.world_size * args.batch_size
-    args.compute_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    args.compute_dtype = (torch.float32 if args.dtype == "fp32"
+                          else torch.bfloat16)
+    args.fp8 = args.dtype == "fp8"
     os.makedirs(args.output_dir, exist_ok=True)
 
     gan = CycleGAN(args, ctx)
@@ -84,8 +86,10 @@ def main():
     elapsed = time.perf_counter() - t0
     ctx.barrier()
 
-    # MAX elapsed over ranks -> whole-job throughput
-    t = torch.tensor([elapsed], dtype=torch.float64)
+    # MAX elapsed over ranks -> whole-job throughput (NCCL needs a device
+    # tensor)
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=ctx.device if on_gpu else "cpu")
     if ctx.distributed:
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
     elapsed = t.item()

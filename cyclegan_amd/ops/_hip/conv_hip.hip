@@ -39,6 +39,7 @@ struct ConvParams {
   int stride, pt, pl;
   int reflect;                   // conv_fwd only
   int act;  float slope;
+  const float* dq;               // fp8 path: [0] = 1/(sx*sw)
   long M, KTOT;
   int mtiles, ntiles;
 };
@@ -446,6 +447,242 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
         int rl = wm0 + mf * 16 + fg * 4 + r;
         if (!sm.rowok[rl]) continue;
         float v = apply_act(acc[mf][nf][r] + bv, p.act, p.slope);
+        p.y[sm.rowyb[rl] + n] = f2b(v);
+      }
+    }
+  }
+}
+
+// ---------------- fp8 (e4m3) forward conv ----------------
+// CDNA4 fp8 MFMA path (BASELINE config 5): activations and weights are
+// quantized to OCP e4m3 with per-tensor scales; fp32 accumulate; the
+// epilogue multiplies by 1/(sx*sw) before bias/activation. Byte-level tile
+// geometry is IDENTICAL to the bf16 glds kernel (128-B rows, 16-B lane
+// granules, same XOR swizzle) — a K-step covers 128 fp8 elements and runs
+// 4 mfma_f32_16x16x32_fp8_fp8 substeps. Forward only: backward reuses the
+// bf16 kernels from the saved bf16 activations (master-grad numerics).
+
+#include <hip/hip_fp8.h>
+
+constexpr int BKF = 128;  // fp8 K-step (elements == bytes)
+
+struct Fp8Smem {
+  unsigned char A[2][BM * BKF];
+  unsigned char Bt[2][BN * BKF];
+  long rowyb[BM];
+  int rowih[BM], rowiw[BM];
+  unsigned rowxb[BM];  // byte offsets (1 B / element)
+  char rowok[BM];
+};
+
+__global__ void quant_fp8_kernel(const short* __restrict__ x,
+                                 const float* __restrict__ scale,
+                                 unsigned char* __restrict__ y, long n) {
+  long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  float s = scale[0];
+  if (i + 8 <= n) {
+    v8s v = *(const v8s*)(x + i);
+    unsigned char out[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      __hip_fp8_e4m3 q(b2f(v[j]) * s);
+      out[j] = q.__x;
+    }
+    *(uint2*)(y + i) = *(uint2*)out;
+  } else if (i < n) {
+    for (long k = i; k < n; ++k) {
+      __hip_fp8_e4m3 q(b2f(x[k]) * scale[0]);
+      y[k] = q.__x;
+    }
+  }
+}
+
+template <int STRIDE>
+__global__ __launch_bounds__(NTHREADS) void conv_fp8_kernel(ConvParams p) {
+  const int stride = STRIDE ? STRIDE : p.stride;
+  __shared__ Fp8Smem sm;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int bid = blockIdx.x;
+  const int mt = bid % p.mtiles, nt = bid / p.mtiles;
+  const long m0 = (long)mt * BM;
+  const int n0 = nt * BN;
+  const bool reflect = p.reflect != 0;
+
+  for (int r = tid; r < BM; r += NTHREADS) {
+    long m = m0 + r;
+    bool ok = m < p.M;
+    long mm = ok ? m : 0;
+    int ow = (int)(mm % p.OW);
+    int oh = (int)((mm / p.OW) % p.OH);
+    int b = (int)(mm / ((long)p.OW * p.OH));
+    sm.rowok[r] = ok;
+    sm.rowxb[r] = (unsigned)((long)b * p.H * p.W * p.Cin);
+    sm.rowyb[r] = ((long)(b * p.OH + oh) * p.OW + ow) * p.Cout;
+    sm.rowih[r] = oh * stride - p.pt;
+    sm.rowiw[r] = ow * stride - p.pl;
+  }
+  __syncthreads();
+
+  const int lr = lane >> 3;
+  const int klog = ((lane & 7) ^ lr) << 4;  // 16-elem granule
+  int aih[4], aiw[4];
+  unsigned axb[4];
+  bool aok[4];
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int r = w * 32 + j * 8 + lr;
+    aih[j] = sm.rowih[r];
+    aiw[j] = sm.rowiw[r];
+    axb[j] = sm.rowxb[r];
+    aok[j] = sm.rowok[r];
+  }
+  const int bn[2] = {(w * 2 + 0) * 8 + lr + n0, (w * 2 + 1) * 8 + lr + n0};
+
+  auto rx = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)p.x, 0, (unsigned)((long)p.B * p.H * p.W * p.Cin), 0x00020000);
+  auto rw = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)p.w, 0, (unsigned)(p.Cout * p.KTOT), 0x00020000);
+
+  const int nk = (int)((p.KTOT + BKF - 1) / BKF);
+  const bool big_ci = p.Cin >= BKF;
+
+  long kcur;
+  bool kv;
+  int ci, dkh, dkw;
+  unsigned avo[4], bvo[2];
+  bool avalid[4], bnv[2];
+
+  auto recompute_a = [&]() {
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      avalid[j] = false;
+      avo[j] = 0xFF000000u;
+      if (!aok[j]) continue;
+      bool valid = true;
+      int ih = aih[j] + dkh, iw = aiw[j] + dkw;
+      if (reflect) {
+        ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
+      } else {
+        valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+      }
+      if (valid) {
+        avalid[j] = true;
+        avo[j] = axb[j] + (unsigned)(((long)ih * p.W + iw) * p.Cin + ci);
+      }
+    }
+  };
+
+  auto init_state = [&]() {
+    kcur = klog;
+    kv = kcur < p.KTOT;
+    int tap = (int)(kcur / p.Cin);
+    ci = (int)(kcur - (long)tap * p.Cin);
+    dkh = tap / p.KW;
+    dkw = tap - dkh * p.KW;
+    recompute_a();
+    #pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      bnv[j] = bn[j] < p.Cout;
+      bvo[j] = (unsigned)((long)bn[j] * p.KTOT + kcur);
+    }
+  };
+
+  auto advance = [&]() {
+    kcur += BKF;
+    kv = kcur < p.KTOT;
+    #pragma unroll
+    for (int j = 0; j < 2; ++j) bvo[j] += BKF;
+    if (big_ci) {
+      ci += BKF;
+      if (ci >= p.Cin) {
+        ci -= p.Cin;
+        if (++dkw == p.KW) { dkw = 0; ++dkh; }
+        recompute_a();
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) avo[j] += BKF;
+      }
+    } else {
+      int tap = (int)(kcur / p.Cin);
+      ci = (int)(kcur - (long)tap * p.Cin);
+      dkh = tap / p.KW;
+      dkw = tap - dkh * p.KW;
+      recompute_a();
+    }
+  };
+
+  auto stage = [&](int buf) {
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      unsigned vo = (kv && avalid[j]) ? avo[j] : 0xFF000000u;
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * 4 + j) * 1024],
+          16, vo, 0, 0, 0);
+    }
+    #pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      unsigned vo = (kv && bnv[j]) ? bvo[j] : 0xFF000000u;
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rw, (__attribute__((address_space(3))) void*)&sm.Bt[buf][(w * 2 + j) * 1024],
+          16, vo, 0, 0, 0);
+    }
+  };
+
+  v4f acc[4][2] = {};
+  const int wr = w >> 1, wc = w & 1;
+  const int wm0 = wr * 64, wn0 = wc * 32;
+  const int fr = lane & 15;
+  const int fg = lane >> 4;
+  const int swz = (fr & 7) << 4;
+
+  init_state();
+  stage(0);
+  __syncthreads();
+
+  for (int kt = 0; kt < nk; ++kt) {
+    if (kt + 1 < nk) {
+      advance();
+      stage((kt + 1) & 1);
+    }
+    const char* Ab = (const char*)sm.A[kt & 1];
+    const char* Bb = (const char*)sm.Bt[kt & 1];
+    #pragma unroll
+    for (int kk = 0; kk < BKF; kk += 32) {
+      const int kbyte = kk + fg * 8;
+      long a0 = *(const long*)(Ab + ((wm0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
+      long a1 = *(const long*)(Ab + ((wm0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
+      long a2 = *(const long*)(Ab + ((wm0 + 2 * 16 + fr) << 7) + (kbyte ^ swz));
+      long a3 = *(const long*)(Ab + ((wm0 + 3 * 16 + fr) << 7) + (kbyte ^ swz));
+      long b0 = *(const long*)(Bb + ((wn0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
+      long b1 = *(const long*)(Bb + ((wn0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a1, b1, acc[1][1], 0, 0, 0);
+      acc[2][0] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a2, b0, acc[2][0], 0, 0, 0);
+      acc[2][1] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a2, b1, acc[2][1], 0, 0, 0);
+      acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a3, b0, acc[3][0], 0, 0, 0);
+      acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a3, b1, acc[3][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const float dqs = p.dq ? p.dq[0] : 1.f;
+  #pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+    int n = n0 + wn0 + nf * 16 + fr;
+    if (n >= p.Cout) continue;
+    float bv = p.bias ? b2f(p.bias[n]) : 0.f;
+    #pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int rl = wm0 + mf * 16 + fg * 4 + r;
+        if (!sm.rowok[rl]) continue;
+        float v = apply_act(acc[mf][nf][r] * dqs + bv, p.act, p.slope);
         p.y[sm.rowyb[rl] + n] = f2b(v);
       }
     }
@@ -1483,6 +1720,63 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
   p.dw = (float*)dw.mutable_data_ptr();
   hipLaunchKernelGGL(wgrad_kernel, grid, dim3(NTHREADS), 0, stream, p);
   return dw;
+}
+
+// ---- fp8 host wrappers ----
+at::Tensor quant_fp8(at::Tensor x, at::Tensor scale) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.is_contiguous());
+  TORCH_CHECK(scale.scalar_type() == at::kFloat);
+  auto y = at::empty_like(x, x.options().dtype(at::kByte));
+  long n = x.numel();
+  long blocks = (n / 8 + 255) / 256 + 1;
+  hipLaunchKernelGGL(quant_fp8_kernel, dim3((unsigned)blocks), dim3(256), 0,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     (const short*)x.const_data_ptr(),
+                     (const float*)scale.const_data_ptr(),
+                     (unsigned char*)y.mutable_data_ptr(), n);
+  return y;
+}
+
+at::Tensor conv2d_fp8_fwd(at::Tensor xq, at::Tensor wq, at::Tensor dq,
+                          c10::optional<at::Tensor> bias, int64_t stride,
+                          int64_t pt, int64_t pb, int64_t pl, int64_t pr,
+                          bool reflect, int64_t act, double slope) {
+  TORCH_CHECK(xq.is_cuda() && xq.scalar_type() == at::kByte && xq.is_contiguous());
+  TORCH_CHECK(wq.scalar_type() == at::kByte && wq.is_contiguous());
+  TORCH_CHECK(xq.size(3) == wq.size(3) && xq.size(3) % 16 == 0,
+              "fp8 conv: Cin % 16 != 0");
+  int H = xq.size(1), W = xq.size(2);
+  int KH = wq.size(1), KW = wq.size(2);
+  int OH = (H + pt + pb - KH) / stride + 1;
+  int OW = (W + pl + pr - KW) / stride + 1;
+  auto y = at::empty({xq.size(0), OH, OW, wq.size(0)},
+                     xq.options().dtype(at::kBFloat16));
+  ConvParams p{};
+  p.x = (const short*)xq.const_data_ptr();
+  p.w = (const short*)wq.const_data_ptr();
+  p.bias = bias.has_value() ? (const short*)bias->const_data_ptr() : nullptr;
+  p.dq = (const float*)dq.const_data_ptr();
+  p.y = (short*)y.mutable_data_ptr();
+  p.B = xq.size(0); p.H = H; p.W = W; p.Cin = xq.size(3);
+  p.OH = OH; p.OW = OW; p.Cout = wq.size(0);
+  p.KH = KH; p.KW = KW;
+  p.stride = stride; p.pt = pt; p.pl = pl;
+  p.reflect = reflect ? 1 : 0;
+  p.act = act; p.slope = (float)slope;
+  p.M = (long)p.B * OH * OW;
+  p.KTOT = (long)KH * KW * p.Cin;
+  p.mtiles = cdiv(p.M, BM);
+  p.ntiles = cdiv(p.Cout, BN);
+  TORCH_CHECK((long)p.B * H * W * p.Cin < (1L << 31) &&
+              (long)p.Cout * p.KTOT < (1L << 31));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  dim3 grid(p.mtiles * p.ntiles);
+  switch (p.stride) {
+    case 1: hipLaunchKernelGGL((conv_fp8_kernel<1>), grid, dim3(NTHREADS), 0, stream, p); break;
+    case 2: hipLaunchKernelGGL((conv_fp8_kernel<2>), grid, dim3(NTHREADS), 0, stream, p); break;
+    default: hipLaunchKernelGGL((conv_fp8_kernel<0>), grid, dim3(NTHREADS), 0, stream, p);
+  }
+  return y;
 }
 
 // ---- glds probe (test aid): verify raw_ptr_buffer_load_lds semantics ----

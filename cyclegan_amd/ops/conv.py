@@ -34,7 +34,7 @@ import torch.nn.functional as F
 
 from . import backend
 from .shadow import (compute_weight, compute_weight_t, compute_weight_p,
-                     compute_weight_tp)
+                     compute_weight_tp, fp8_weight_shadow)
 
 ACT_NONE, ACT_RELU, ACT_LRELU, ACT_TANH = 0, 1, 2, 3
 
@@ -158,6 +158,46 @@ class _ConvFn(torch.autograd.Function):
         return dx, dw, db, None, None, None, None, None
 
 
+_FP8_MODE = False
+
+
+def set_fp8_mode(on: bool):
+    """Enable the CDNA4 fp8 (e4m3) forward-conv path (BASELINE config 5):
+    per-tensor-scaled fp8 weights+activations on the fp8 MFMA, fp32
+    accumulate, bf16 outputs; backward runs on the saved bf16 activations."""
+    global _FP8_MODE
+    _FP8_MODE = bool(on)
+
+
+def fp8_mode() -> bool:
+    return _FP8_MODE
+
+
+class _ConvFp8Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, pads, reflect, act, slope):
+        ext = backend.ext()
+        cout = w.shape[0]
+        xp = _pad_channels(x).contiguous()
+        wq, sw = fp8_weight_shadow(w)
+        amax = xp.detach().abs().amax().float().clamp(min=1e-12)
+        sx = (448.0 / amax).clamp(max=65504.0)
+        xq = ext.quant_fp8(xp, sx)
+        dq = (sx * sw).reciprocal()
+        bc = compute_weight(bias, x) if bias is not None else None
+        if bc is not None and cout < 8:
+            bc = _pad_channels(bc.view(1, 1, 1, -1)).view(-1)
+        y = ext.conv2d_fp8_fwd(xq, wq, dq, bc, stride, *pads, reflect, act, slope)
+        if cout < 8:
+            y = y[..., :cout].contiguous()
+        ctx.save_for_backward(xp, w, y)
+        ctx.conf = (stride, pads, reflect, act, slope, bias is not None,
+                    x.shape[3])
+        return y
+
+    backward = _ConvFn.backward
+
+
 class _ConvTFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, bias, stride, pt, pl, out_h, out_w, act, slope):
@@ -216,6 +256,10 @@ def conv2d(
         pads = tuple(padding)
     a = _ACT[act]
     if backend.use_hip(x, w):
+        if (_FP8_MODE and x.dtype == torch.bfloat16
+                and max(w.shape[3], 8) % 16 == 0):
+            return _ConvFp8Fn.apply(x, w, bias, stride, pads,
+                                    pad_mode == "reflect", a, slope)
         return _ConvFn.apply(x, w, bias, stride, pads, pad_mode == "reflect", a, slope)
     wc = w if w.dtype == x.dtype else w.to(x.dtype)
     bc = bias if (bias is None or bias.dtype == x.dtype) else bias.to(x.dtype)

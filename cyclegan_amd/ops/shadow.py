@@ -17,6 +17,7 @@ _cache: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
 _cache_t: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
 _cache_p: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
 _cache_tp: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
+_cache_f8: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
 
 
 def bf16_shadow(t: torch.Tensor) -> torch.Tensor:
@@ -102,3 +103,19 @@ def compute_weight_t(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
         wt = wt.to(like.dtype)
     _cache_t[w] = (ver, like.dtype, wt)
     return wt
+
+
+def fp8_weight_shadow(w: torch.Tensor):
+    """(wq uint8 e4m3 padded OHWI, sw fp32 scale tensor), cached per master
+    version. scale = 448/amax (OCP e4m3 max normal)."""
+    from . import backend
+    ent = _cache_f8.get(w)
+    ver = w._version
+    if ent is not None and ent[0] == ver:
+        return ent[1], ent[2]
+    wp = _pad_dims(w.detach()).contiguous()
+    amax = wp.abs().amax().float().clamp(min=1e-12)
+    sw = (448.0 / amax).clamp(max=65504.0)
+    wq = backend.ext().quant_fp8(wp.to(torch.bfloat16).contiguous(), sw)
+    _cache_f8[w] = (ver, wq, sw)
+    return wq, sw

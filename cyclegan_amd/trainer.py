@@ -42,6 +42,10 @@ class CycleGAN:
         self.global_batch_size = args.global_batch_size
         self.compute_dtype = getattr(args, "compute_dtype", None) or (
             torch.bfloat16 if self.device.type == "cuda" else torch.float32)
+        if getattr(args, "fp8", False):
+            from .ops import set_fp8_mode
+            set_fp8_mode(True)
+            self.compute_dtype = torch.bfloat16
 
         self.checkpoint_dir = os.path.join(args.output_dir, "checkpoints")
         if ctx.is_main:

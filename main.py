@@ -68,7 +68,9 @@ def main(args):
 
     args.global_batch_size = ctx.world_size * args.batch_size
     if args.dtype:
-        args.compute_dtype = {"fp32": torch.float32, "bf16": torch.bfloat16}[args.dtype]
+        args.compute_dtype = {"fp32": torch.float32, "bf16": torch.bfloat16,
+                              "fp8": torch.bfloat16}[args.dtype]
+        args.fp8 = args.dtype == "fp8"
     else:
         args.compute_dtype = None
     print(f"Number of devices: {ctx.world_size}")
@@ -122,7 +124,7 @@ if __name__ == "__main__":
                              "default: synthetic horse2zebra-shaped data")
     parser.add_argument("--image_size", default=256, type=int)
     parser.add_argument("--num_residual_blocks", default=9, type=int)
-    parser.add_argument("--dtype", default=None, choices=[None, "fp32", "bf16"])
+    parser.add_argument("--dtype", default=None, choices=[None, "fp32", "bf16", "fp8"])
     parser.add_argument("--num_train_samples", default=None, type=int)
     parser.add_argument("--num_test_samples", default=None, type=int)
     parser.add_argument("--seed", default=1234, type=int)

@@ -1217,6 +1217,11 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
 
 typedef bf16r v4bfx __attribute__((ext_vector_type(4)));
 
+// XOR swizzles for the wgrad images. Row bit 3 feeds cb bit 4 so the two
+// 16-lane tr groups of a 32-lane bank half land on different 16-B slots.
+#define AXOR(row) ((((row) & 7) << 5) | ((((row) >> 3) & 1) << 4))
+#define DXOR(row) ((((row) & 3) << 5) | ((((row) >> 3) & 1) << 4))
+
 DEV v4bfx tr16_read(const char* plds) {
   return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
       (__attribute__((address_space(3))) v4bfx*)plds);
@@ -1256,7 +1261,7 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
   #pragma unroll
   for (int j = 0; j < 4; ++j) {
     int row = (w * 4 + j) * 4 + (lane >> 4);
-    int cb = ((lane & 15) * 16) ^ ((row & 7) << 5);
+    int cb = ((lane & 15) * 16) ^ AXOR(row);
     long k = k0 + cb / 2;
     a_row[j] = row;
     a_kv[j] = k < p.KTOT;
@@ -1272,7 +1277,7 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
   #pragma unroll
   for (int j = 0; j < 2; ++j) {
     int row = (w * 2 + j) * 8 + (lane >> 3);
-    int cb = ((lane & 7) * 16) ^ ((row & 3) << 5);
+    int cb = ((lane & 7) * 16) ^ DXOR(row);
     d_row[j] = row;
     d_n[j] = n0 + cb / 2;
   }
@@ -1359,9 +1364,9 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
       for (int kf = 0; kf < 4; ++kf) {
         int cbl = (wk0 + kf * 16) * 2 + tr_cb_a;
         int row = kk + fg * 8 + tr_row_a;
-        v4bfx lo = tr16_read(Ab + row * 256 + (cbl ^ ((row & 7) << 5)));
+        v4bfx lo = tr16_read(Ab + row * 256 + (cbl ^ AXOR(row)));
         row += 4;
-        v4bfx hi = tr16_read(Ab + row * 256 + (cbl ^ ((row & 7) << 5)));
+        v4bfx hi = tr16_read(Ab + row * 256 + (cbl ^ AXOR(row)));
         a[kf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
       v8bf bfr[2];
@@ -1369,9 +1374,9 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
       for (int nf = 0; nf < 2; ++nf) {
         int cbl = (wn0 + nf * 16) * 2 + tr_cb_a;
         int row = kk + fg * 8 + tr_row_a;
-        v4bfx lo = tr16_read(Db + row * 128 + (cbl ^ ((row & 3) << 5)));
+        v4bfx lo = tr16_read(Db + row * 128 + (cbl ^ DXOR(row)));
         row += 4;
-        v4bfx hi = tr16_read(Db + row * 128 + (cbl ^ ((row & 3) << 5)));
+        v4bfx hi = tr16_read(Db + row * 128 + (cbl ^ DXOR(row)));
         bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
       acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[0], bfr[0], acc[0][0], 0, 0, 0);

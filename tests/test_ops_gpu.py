@@ -283,3 +283,33 @@ def test_fp8_train_step(tmp_path):
             assert torch.isfinite(v), k
     finally:
         convmod.set_fp8_mode(False)
+
+
+@pytest.mark.parametrize("C", [64, 256, 512])
+def test_instnorm_channel_widths(C):
+    B, H, W = 2, 12, 12
+    x = mk((B, H, W, C), seed=40 + C)
+    g = torch.randn(C, generator=torch.Generator().manual_seed(41)).to(DEV) * 0.5
+    bt = torch.randn(C, generator=torch.Generator().manual_seed(42)).to(DEV) * 0.1
+    y = ops.instance_norm(x, g, bt, act="relu")
+    ref = _in_ref(x.float().cpu(), g.float().cpu(), bt.float().cpu(), 1e-3, 1, 0.2, None)
+    check(y, ref, 0.05, f"in_c{C}")
+
+
+def test_conv_batch8_and_odd_spatial():
+    # batched-call shapes (B=8) and non-power-of-two spatial
+    x = mk((8, 24, 24, 64), seed=50)
+    w = mk((128, 3, 3, 64), seed=51, scale=0.2)
+    y = ops.conv2d(x, w, None, 2, "same", "zeros")
+    ref = _conv_ref(x.float().cpu(), w.float().cpu(), None, 2,
+                    same_pads(24, 24, 3, 3, 2), "zeros")
+    check(y, ref, 0.05, "b8odd")
+
+
+def test_conv_512_spatial_shape():
+    x = mk((1, 128, 128, 64), seed=52)
+    w = mk((64, 7, 7, 64), seed=53, scale=0.1)
+    y = ops.conv2d(x, w, None, 1, (3, 3, 3, 3), "reflect")
+    ref = _conv_ref(x.float().cpu(), w.float().cpu(), None, 1,
+                    (3, 3, 3, 3), "reflect")
+    check(y, ref, 0.05, "large_spatial")

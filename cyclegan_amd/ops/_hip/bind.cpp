@@ -24,7 +24,9 @@ std::vector<at::Tensor> instnorm_fwd(at::Tensor, at::Tensor, at::Tensor,
                                      double, int64_t, double,
                                      c10::optional<at::Tensor>);
 std::vector<at::Tensor> instnorm_bwd(at::Tensor, at::Tensor, at::Tensor,
-                                     at::Tensor, at::Tensor);
+                                     at::Tensor, at::Tensor,
+                                     c10::optional<at::Tensor>, int64_t,
+                                     double);
 at::Tensor act_bwd(at::Tensor, at::Tensor, int64_t, double);
 at::Tensor reflect_pad_fwd(at::Tensor, int64_t, int64_t, int64_t, int64_t);
 at::Tensor reflect_pad_bwd(at::Tensor, int64_t, int64_t, int64_t, int64_t);

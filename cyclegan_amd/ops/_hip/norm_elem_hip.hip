@@ -143,11 +143,19 @@ __global__ __launch_bounds__(NT) void in_norm_kernel(
 }
 
 // ---- backward pass 1: slab partials of s1 = Σ dy, s2 = Σ dy*xhat ----
+DEV float actbw(float d, float yy, int act, float slope) {
+  if (act == ACT_RELU) return yy > 0.f ? d : 0.f;
+  if (act == ACT_LRELU) return yy > 0.f ? d : d * slope;
+  if (act == ACT_TANH) return d * (1.f - yy * yy);
+  return d;
+}
+
 __global__ __launch_bounds__(NT) void in_bwd_reduce_kernel(
     const short* __restrict__ dy, const short* __restrict__ x,
-    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const short* __restrict__ yact, const float* __restrict__ mean,
+    const float* __restrict__ rstd,
     float* __restrict__ p1, float* __restrict__ p2, int B, long HW, int C,
-    int S) {
+    int S, int act, float slope) {
   int b = blockIdx.x / S;
   int sl = blockIdx.x % S;
   long rows = (HW + S - 1) / S;
@@ -170,9 +178,12 @@ __global__ __launch_bounds__(NT) void in_bwd_reduce_kernel(
     long off = base + r * C + g * 8;
     v8s dv = *(const v8s*)(dy + off);
     v8s xv = *(const v8s*)(x + off);
+    v8s yv = {};
+    if (act != ACT_NONE) yv = *(const v8s*)(yact + off);
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float d = b2f(dv[j]);
+      if (act != ACT_NONE) d = actbw(d, b2f(yv[j]), act, slope);
       float xh = (b2f(xv[j]) - mv[j]) * rv[j];
       a1[j] += d; a2[j] += d * xh;
     }
@@ -201,10 +212,11 @@ __global__ __launch_bounds__(NT) void in_bwd_reduce_kernel(
 // ---- backward pass 2: dx; also dgamma/dbeta reduce over b ----
 __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
     const short* __restrict__ dy, const short* __restrict__ x,
-    const float* __restrict__ gamma, const float* __restrict__ mean,
+    const short* __restrict__ yact, const float* __restrict__ gamma,
+    const float* __restrict__ mean,
     const float* __restrict__ rstd, const float* __restrict__ p1,
     const float* __restrict__ p2, short* __restrict__ dx, int B, long HW,
-    int C, int S) {
+    int C, int S, int act, float slope) {
   int b = blockIdx.x / S;
   int sl = blockIdx.x % S;
   long rows = (HW + S - 1) / S;
@@ -250,6 +262,8 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
       long off = base + r * C + g * 8;
       v8s dv = *(const v8s*)(dy + off);
       v8s xv = *(const v8s*)(x + off);
+      v8s yv = {};
+      if (act != ACT_NONE) yv = *(const v8s*)(yact + off);
       v8s out;
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -257,6 +271,7 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
         float rs = srstd[c];
         float xh = (b2f(xv[j]) - smean[c]) * rs;
         float d = b2f(dv[j]);
+        if (act != ACT_NONE) d = actbw(d, b2f(yv[j]), act, slope);
         out[j] = f2b(gamma[c] * rs * (d - sm1[c] - xh * sm2[c]));
       }
       *(v8s*)(dx + off) = out;
@@ -497,7 +512,9 @@ std::vector<at::Tensor> instnorm_fwd(at::Tensor x, at::Tensor gamma,
 
 std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
                                      at::Tensor gamma, at::Tensor mean,
-                                     at::Tensor rstd) {
+                                     at::Tensor rstd,
+                                     c10::optional<at::Tensor> yact,
+                                     int64_t act, double slope) {
   TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16 &&
               dy.is_contiguous() && x.is_contiguous());
   int B = x.size(0), C = x.size(3);
@@ -513,22 +530,26 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
   auto dgb = at::zeros({2, C}, fopt);
   auto dbeta = dgb[0];
   auto dgamma = dgb[1];
+  const short* yp = yact.has_value()
+                        ? (const short*)yact->const_data_ptr() : nullptr;
   hipLaunchKernelGGL(in_bwd_reduce_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)dy.const_data_ptr(),
-                     (const short*)x.const_data_ptr(),
+                     (const short*)x.const_data_ptr(), yp,
                      (const float*)mean.const_data_ptr(),
                      (const float*)rstd.const_data_ptr(),
                      (float*)p1.mutable_data_ptr(),
-                     (float*)p2.mutable_data_ptr(), B, HW, C, S);
+                     (float*)p2.mutable_data_ptr(), B, HW, C, S, (int)act,
+                     (float)slope);
   hipLaunchKernelGGL(in_bwd_dx_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)dy.const_data_ptr(),
-                     (const short*)x.const_data_ptr(),
+                     (const short*)x.const_data_ptr(), yp,
                      (const float*)gamma.const_data_ptr(),
                      (const float*)mean.const_data_ptr(),
                      (const float*)rstd.const_data_ptr(),
                      (const float*)p1.const_data_ptr(),
                      (const float*)p2.const_data_ptr(),
-                     (short*)dx.mutable_data_ptr(), B, HW, C, S);
+                     (short*)dx.mutable_data_ptr(), B, HW, C, S, (int)act,
+                     (float)slope);
   hipLaunchKernelGGL(in_bwd_dgb_kernel,
                      dim3(cdiv64(C, 256), cdiv64(S * B, 8)), dim3(256), 0,
                      stream, (const float*)p1.const_data_ptr(),

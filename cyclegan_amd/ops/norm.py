@@ -56,11 +56,17 @@ class _InstNormFn(torch.autograd.Function):
         eps, act, slope, has_res = ctx.conf
         ext = backend.ext()
         dy = dy.contiguous()
-        if act != ACT_NONE:
+        if has_res and act != ACT_NONE:
+            # rare combo: residual grad needs the transformed dy explicitly
             dy = ext.act_bwd(dy, y, act, slope)
-        # residual grad: identity (post-norm add)
-        dres = dy if has_res else None
-        dx, dgamma, dbeta = ext.instnorm_bwd(dy, x, gamma.float(), mean, rstd)
+            dres = dy
+            dx, dgamma, dbeta = ext.instnorm_bwd(dy, x, gamma.float(), mean,
+                                                 rstd, None, ACT_NONE, 0.0)
+        else:
+            dres = dy if has_res else None
+            dx, dgamma, dbeta = ext.instnorm_bwd(
+                dy, x, gamma.float(), mean, rstd,
+                y if act != ACT_NONE else None, act, slope)
         return (dx, dgamma.to(gamma.dtype), dbeta, None, None, None, dres)
 
 

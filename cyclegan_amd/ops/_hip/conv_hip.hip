@@ -1283,9 +1283,30 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
     d_n[j] = n0 + cb / 2;
   }
 
-  // incremental m-decode (ow/oh/b advance by +WG_BM each m-step; no
-  // per-step division — the wrap loop runs <= WG_BM/OW + 1 iterations)
-  int a_ow[4], a_oh[4], a_b[4];
+  // incremental m-decode AND incremental voffsets: interior steps cost two
+  // small multiplies per instruction; full address recomputation happens
+  // only at image borders / batch wraps (O(1/OH) of steps).
+  const unsigned C2 = (unsigned)p.Cin * 2;
+  const unsigned WC2 = (unsigned)p.W * C2;
+  int a_ow[4], a_oh[4], a_b[4], a_ih[4], a_iw[4];
+  unsigned avo[4];
+  bool ainb[4], avalid[4];
+  unsigned dvo[2];
+
+  auto full_a = [&](int j) {
+    int ih = a_ih[j], iw = a_iw[j];
+    bool valid = true;
+    if (p.reflect) {
+      ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
+    } else {
+      valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+    }
+    avalid[j] = valid;
+    avo[j] = valid ? (unsigned)(((((long)a_b[j] * p.H + ih) * p.W + iw) *
+                                 p.Cin + a_ci[j]) * 2)
+                   : 0xFF000000u;
+  };
+
   #pragma unroll
   for (int j = 0; j < 4; ++j) {
     long m = mstart + a_row[j];
@@ -1293,44 +1314,53 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
     int t = (int)(m / p.OW);
     a_oh[j] = t % p.OH;
     a_b[j] = t / p.OH;
+    a_ih[j] = a_oh[j] * p.stride - p.pt + a_dkh[j];
+    a_iw[j] = a_ow[j] * p.stride - p.pl + a_dkw[j];
+    ainb[j] = (unsigned)a_ih[j] < (unsigned)p.H &&
+              (unsigned)a_iw[j] < (unsigned)p.W;
+    full_a(j);
   }
+  #pragma unroll
+  for (int j = 0; j < 2; ++j)
+    dvo[j] = (unsigned)(((mstart + d_row[j]) * p.Cout + d_n[j]) * 2);
 
   auto advance = [&]() {
     #pragma unroll
     for (int j = 0; j < 4; ++j) {
+      int ow0 = a_ow[j], oh0 = a_oh[j], b0 = a_b[j];
       a_ow[j] += WG_BM;
       while (a_ow[j] >= p.OW) { a_ow[j] -= p.OW; ++a_oh[j]; }
       while (a_oh[j] >= p.OH) { a_oh[j] -= p.OH; ++a_b[j]; }
+      int dih = (a_oh[j] - oh0) * p.stride;
+      int diw = (a_ow[j] - ow0) * p.stride;
+      a_ih[j] += dih;
+      a_iw[j] += diw;
+      bool inb = (unsigned)a_ih[j] < (unsigned)p.H &&
+                 (unsigned)a_iw[j] < (unsigned)p.W;
+      if (a_b[j] == b0 && inb && ainb[j]) {
+        avo[j] += (unsigned)(dih * (int)WC2 + diw * (int)C2);
+      } else {
+        full_a(j);
+      }
+      ainb[j] = inb;
     }
+    #pragma unroll
+    for (int j = 0; j < 2; ++j) dvo[j] += (unsigned)(WG_BM * p.Cout * 2);
   };
 
   auto stage = [&](int buf, long ms) {
     #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      unsigned vo = 0xFFFFFF00u;
-      long m = ms + a_row[j];
-      if (a_kv[j] && m < p.M) {
-        int ih = a_oh[j] * p.stride - p.pt + a_dkh[j];
-        int iw = a_ow[j] * p.stride - p.pl + a_dkw[j];
-        bool valid = true;
-        if (p.reflect) {
-          ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
-        } else {
-          valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
-        }
-        if (valid)
-          vo = (unsigned)(((((long)a_b[j] * p.H + ih) * p.W + iw) * p.Cin + a_ci[j]) * 2);
-      }
+      unsigned vo = (a_kv[j] && avalid[j] && ms + a_row[j] < p.M)
+                        ? avo[j] : 0xFF000000u;
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
           rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * 4 + j) * 512],
           16, vo, 0, 0, 0);
     }
     #pragma unroll
     for (int j = 0; j < 2; ++j) {
-      unsigned vo = 0xFFFFFF00u;
-      long m = ms + d_row[j];
-      if (m < p.M && d_n[j] < p.Cout)
-        vo = (unsigned)((m * p.Cout + d_n[j]) * 2);
+      unsigned vo = (d_n[j] < p.Cout && ms + d_row[j] < p.M)
+                        ? dvo[j] : 0xFF000000u;
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
           rd, (__attribute__((address_space(3))) void*)&sm.D[buf][(w * 2 + j) * 512],
           16, vo, 0, 0, 0);

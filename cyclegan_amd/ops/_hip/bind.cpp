@@ -28,6 +28,7 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor, at::Tensor, at::Tensor,
                                      c10::optional<at::Tensor>, int64_t,
                                      double);
 at::Tensor act_bwd(at::Tensor, at::Tensor, int64_t, double);
+at::Tensor channel_sum(at::Tensor);
 at::Tensor reflect_pad_fwd(at::Tensor, int64_t, int64_t, int64_t, int64_t);
 at::Tensor reflect_pad_bwd(at::Tensor, int64_t, int64_t, int64_t, int64_t);
 at::Tensor persample_loss_fwd(at::Tensor, at::Tensor, bool);
@@ -53,6 +54,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("instnorm_fwd", &cyg::instnorm_fwd);
   m.def("instnorm_bwd", &cyg::instnorm_bwd);
   m.def("act_bwd", &cyg::act_bwd);
+  m.def("channel_sum", &cyg::channel_sum);
   m.def("reflect_pad_fwd", &cyg::reflect_pad_fwd);
   m.def("reflect_pad_bwd", &cyg::reflect_pad_bwd);
   m.def("persample_loss_fwd", &cyg::persample_loss_fwd);

@@ -297,6 +297,36 @@ __global__ void in_bwd_dgb_kernel(const float* __restrict__ p1,
   atomicAdd(&dgamma[c], dg);
 }
 
+// ---- channel sum: out[c] = sum over (b,h,w) of x[...,c] (bias grads) ----
+__global__ __launch_bounds__(NT) void channel_sum_kernel(
+    const short* __restrict__ x, float* __restrict__ out, long rows, int C,
+    int S) {
+  int sl = blockIdx.x;
+  long per = (rows + S - 1) / S;
+  long r0 = sl * per, r1 = min(r0 + per, rows);
+  const int gpr = C / 8;
+  const int tid = threadIdx.x;
+  int g = tid % gpr;
+  int rstep = NT / gpr;
+  int rof = tid / gpr;
+  float a[8] = {};
+  for (long r = r0 + rof; r < r1; r += rstep) {
+    v8s v = *(const v8s*)(x + r * C + g * 8);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) a[j] += b2f(v[j]);
+  }
+  __shared__ float red[NT * 17];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) red[tid * 17 + j] = a[j];
+  __syncthreads();
+  for (int c = tid; c < C; c += NT) {
+    int g2 = c / 8, j = c % 8;
+    float t = 0;
+    for (int k = 0; k < rstep; ++k) t += red[(g2 + k * gpr) * 17 + j];
+    atomicAdd(&out[c], t);
+  }
+}
+
 // ---- activation backward (from output) ----
 __global__ void act_bwd_kernel(const short* __restrict__ dy,
                                const short* __restrict__ y,
@@ -557,6 +587,21 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
                      (float*)dbeta.mutable_data_ptr(),
                      (float*)dgamma.mutable_data_ptr(), S * B, C);
   return {dx, dgamma, dbeta};
+}
+
+at::Tensor channel_sum(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.is_contiguous());
+  int C = x.size(-1);
+  long rows = x.numel() / C;
+  TORCH_CHECK(C % 8 == 0 && (C / 8) <= NT && NT % (C / 8) == 0,
+              "channel_sum: unsupported C ", C);
+  auto out = at::zeros({C}, x.options().dtype(at::kFloat));
+  int S = (int)std::min<long>(512, std::max<long>(1, rows / 64));
+  hipLaunchKernelGGL(channel_sum_kernel, dim3(S), dim3(NT), 0,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     (const short*)x.const_data_ptr(),
+                     (float*)out.mutable_data_ptr(), rows, C, S);
+  return out;
 }
 
 at::Tensor act_bwd(at::Tensor dy, at::Tensor y, int64_t act, double slope) {

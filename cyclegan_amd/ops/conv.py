@@ -137,10 +137,10 @@ class _ConvFn(torch.autograd.Function):
         dy = dy.contiguous()
         if act != ACT_NONE:
             dy = ext.act_bwd(dy, y, act, slope)
+        dyp = _pad_channels(dy).contiguous()
         db = None
         if has_bias and ctx.needs_input_grad[2]:
-            db = dy.float().sum(dim=(0, 1, 2))
-        dyp = _pad_channels(dy).contiguous()
+            db = ext.channel_sum(dyp)[: dy.shape[-1]]
         dx = dw = None
         if ctx.needs_input_grad[0]:
             wt = compute_weight_tp(w, xp)

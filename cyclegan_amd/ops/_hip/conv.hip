@@ -243,8 +243,12 @@ struct ConvSmemT {
 };
 using ConvSmem = ConvSmemT<2>;
 
-template <bool IS_CONVT, int STRIDE, int NBUF = 2>
-__global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
+template <bool IS_CONVT, int STRIDE, int NBUF = 2, int NW = 4>
+__global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
+  constexpr int NT = NW * 64;        // threads
+  constexpr int API = 16 / NW;       // A glds instructions per wave
+  constexpr int BPI = 8 / NW;        // B glds instructions per wave
+  constexpr int MF = (256 / NW) / 16;  // m-fragments per wave
   const int stride = STRIDE ? STRIDE : p.stride;
   __shared__ ConvSmemT<NBUF> sm;
 
@@ -256,7 +260,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
   const long m0 = (long)mt * BM;
   const int n0 = nt * BN;
 
-  for (int r = tid; r < BM; r += NTHREADS) {
+  for (int r = tid; r < BM; r += NT) {
     long m = m0 + r;
     bool ok = m < p.M;
     long mm = ok ? m : 0;
@@ -279,18 +283,20 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
   // per-lane persistent gather state
   const int lr = lane >> 3;                    // row-in-8 of each instr
   const int klog = ((lane & 7) ^ lr) << 3;     // swizzled k-chunk (elems)
-  int aih[4], aiw[4];
-  unsigned axb[4];
-  bool aok[4];
+  int aih[API], aiw[API];
+  unsigned axb[API];
+  bool aok[API];
   #pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    int r = w * 32 + j * 8 + lr;
+  for (int j = 0; j < API; ++j) {
+    int r = (w * API + j) * 8 + lr;
     aih[j] = sm.rowih[r];
     aiw[j] = sm.rowiw[r];
     axb[j] = sm.rowxb[r];
     aok[j] = sm.rowok[r];
   }
-  const int bn[2] = {(w * 2 + 0) * 8 + lr + n0, (w * 2 + 1) * 8 + lr + n0};
+  int bn[BPI];
+  #pragma unroll
+  for (int j = 0; j < BPI; ++j) bn[j] = (w * BPI + j) * 8 + lr + n0;
 
   auto rx = __builtin_amdgcn_make_buffer_rsrc(
       (void*)p.x, 0, (unsigned)((long)p.B * p.H * p.W * p.Cin * 2), 0x00020000);
@@ -305,14 +311,14 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
   long kcur;
   bool kv;
   int ci, dkh, dkw;
-  unsigned avo[4];
-  bool avalid[4];
-  unsigned bvo[2];
-  bool bnv[2];
+  unsigned avo[API];
+  bool avalid[API];
+  unsigned bvo[BPI];
+  bool bnv[BPI];
 
   auto recompute_a = [&]() {
     #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < API; ++j) {
       avalid[j] = false;
       avo[j] = 0xFF000000u;
       if (!aok[j]) continue;
@@ -347,7 +353,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
     dkw = tap - dkh * p.KW;
     recompute_a();
     #pragma unroll
-    for (int j = 0; j < 2; ++j) {
+    for (int j = 0; j < BPI; ++j) {
       bnv[j] = bn[j] < p.Cout;
       bvo[j] = (unsigned)(((long)bn[j] * p.KTOT + kcur) * 2);
     }
@@ -357,7 +363,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
     kcur += BK;
     kv = kcur < p.KTOT;
     #pragma unroll
-    for (int j = 0; j < 2; ++j) bvo[j] += BK * 2;
+    for (int j = 0; j < BPI; ++j) bvo[j] += BK * 2;
     if (big_ci) {
       ci += BK;
       if (ci >= p.Cin) {
@@ -366,7 +372,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
         recompute_a();
       } else {
         #pragma unroll
-        for (int j = 0; j < 4; ++j) avo[j] += BK * 2;
+        for (int j = 0; j < API; ++j) avo[j] += BK * 2;
       }
     } else {
       // Cin < 64: a K-step crosses several taps — full recompute
@@ -380,24 +386,24 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
 
   auto stage = [&](int buf) {
     #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < API; ++j) {
       unsigned vo = (kv && avalid[j]) ? avo[j] : 0xFF000000u;
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
-          rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * 4 + j) * 512],
+          rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * API + j) * 512],
           16, vo, 0, 0, 0);
     }
     #pragma unroll
-    for (int j = 0; j < 2; ++j) {
+    for (int j = 0; j < BPI; ++j) {
       unsigned vo = (kv && bnv[j]) ? bvo[j] : 0xFF000000u;
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
-          rw, (__attribute__((address_space(3))) void*)&sm.Bt[buf][(w * 2 + j) * 512],
+          rw, (__attribute__((address_space(3))) void*)&sm.Bt[buf][(w * BPI + j) * 512],
           16, vo, 0, 0, 0);
     }
   };
 
-  v4f acc[4][2] = {};
+  v4f acc[MF][2] = {};
   const int wr = w >> 1, wc = w & 1;
-  const int wm0 = wr * 64, wn0 = wc * 32;
+  const int wm0 = wr * (MF * 16), wn0 = wc * 32;
   const int fr = lane & 15;
   const int fg = lane >> 4;
   const int swz = (fr & 7) << 4;  // read-side XOR (bytes)
@@ -440,7 +446,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_glds_kernel(ConvParams p) {
     if (n >= p.Cout) continue;
     float bv = p.bias ? b2f(p.bias[n]) : 0.f;
     #pragma unroll
-    for (int mf = 0; mf < 4; ++mf) {
+    for (int mf = 0; mf < MF; ++mf) {
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int rl = wm0 + mf * 16 + fg * 4 + r;
@@ -696,6 +702,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_fp8_kernel(ConvParams p) {
 // (K_eff = nvh*nvw*Cin), so every MFMA operates on useful data.
 __global__ __launch_bounds__(NTHREADS) void convt_phased_kernel(ConvParams p) {
   constexpr int NBUF = 2;  // (shares the templated loop body text)
+  constexpr int MF = 4;
   __shared__ ConvSmem sm;
 
   const int tid = threadIdx.x;
@@ -880,20 +887,18 @@ __global__ __launch_bounds__(NTHREADS) void convt_phased_kernel(ConvParams p) {
     #pragma unroll
     for (int kk = 0; kk < BK; kk += 32) {
       const int kbyte = (kk + fg * 8) * 2;
-      v8bf a0 = *(const v8bf*)(Ab + ((wm0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
-      v8bf a1 = *(const v8bf*)(Ab + ((wm0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
-      v8bf a2 = *(const v8bf*)(Ab + ((wm0 + 2 * 16 + fr) << 7) + (kbyte ^ swz));
-      v8bf a3 = *(const v8bf*)(Ab + ((wm0 + 3 * 16 + fr) << 7) + (kbyte ^ swz));
-      v8bf b0 = *(const v8bf*)(Bb + ((wn0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
-      v8bf b1 = *(const v8bf*)(Bb + ((wn0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
-      acc[2][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b0, acc[2][0], 0, 0, 0);
-      acc[2][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b1, acc[2][1], 0, 0, 0);
-      acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b0, acc[3][0], 0, 0, 0);
-      acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b1, acc[3][1], 0, 0, 0);
+      v8bf a[MF], b[2];
+      #pragma unroll
+      for (int mf = 0; mf < MF; ++mf)
+        a[mf] = *(const v8bf*)(Ab + ((wm0 + mf * 16 + fr) << 7) + (kbyte ^ swz));
+      #pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        b[nf] = *(const v8bf*)(Bb + ((wn0 + nf * 16 + fr) << 7) + (kbyte ^ swz));
+      #pragma unroll
+      for (int mf = 0; mf < MF; ++mf) {
+        acc[mf][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mf], b[0], acc[mf][0], 0, 0, 0);
+        acc[mf][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mf], b[1], acc[mf][1], 0, 0, 0);
+      }
     }
     if constexpr (NBUF == 2) __syncthreads();
   }
@@ -1539,8 +1544,32 @@ static bool use_pipe3() {
   return v != 0;
 }
 
+static int conv_nw() {
+  static int v = []() {
+    const char* e = getenv("CYG_CONV_NW");
+    return e ? atoi(e) : 8;
+  }();
+  return v;
+}
+
 template <bool IS_CONVT>
 static void launch_glds_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
+  if (conv_nw() == 8) {
+    switch (p.stride) {
+      case 1:
+        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 1, 2, 8>), grid,
+                           dim3(512), 0, stream, p);
+        return;
+      case 2:
+        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 2, 2, 8>), grid,
+                           dim3(512), 0, stream, p);
+        return;
+      default:
+        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 0, 2, 8>), grid,
+                           dim3(512), 0, stream, p);
+        return;
+    }
+  }
   if (use_pipe3()) {
     switch (p.stride) {
       case 1:

@@ -1238,7 +1238,11 @@ struct WgSmem {
   short D[2][WG_BM * WG_BN];   // [m][n] swizzled
 };
 
-__global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
+template <int NW = 4>
+__global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
+  constexpr int API = 16 / NW;   // A glds instructions per wave
+  constexpr int DPI = 8 / NW;    // D glds instructions per wave
+  constexpr int KF = (256 / NW) / 16;  // k-fragments per wave
   __shared__ WgSmem sm;
 
   const int tid = threadIdx.x;
@@ -1260,13 +1264,12 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
       (void*)p.dy, 0, (unsigned)(p.M * p.Cout * 2), 0x00020000);
 
   // ---- per-lane gather identities ----
-  // A: 16 instrs (4/wave): instr j: row = (w*4+j)*4 + lane/16,
-  //    cb_lin = (lane%16)*16, logical k = (cb_lin ^ ((row&7)<<5))/2 + k0
-  int a_row[4], a_k[4], a_tap[4], a_ci[4], a_dkh[4], a_dkw[4];
-  bool a_kv[4];
+  // A: 16 instrs: instr i covers rows i*4..i*4+3; per-wave share = API
+  int a_row[API], a_k[API], a_tap[API], a_ci[API], a_dkh[API], a_dkw[API];
+  bool a_kv[API];
   #pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    int row = (w * 4 + j) * 4 + (lane >> 4);
+  for (int j = 0; j < API; ++j) {
+    int row = (w * API + j) * 4 + (lane >> 4);
     int cb = ((lane & 15) * 16) ^ AXOR(row);
     long k = k0 + cb / 2;
     a_row[j] = row;
@@ -1277,12 +1280,11 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
     a_dkw[j] = tap - a_dkh[j] * p.KW;
     a_k[j] = (int)(k - k0);
   }
-  // D: 8 instrs (2/wave): instr j: row = (w*2+j)*8 + lane/8,
-  //    cb_lin = (lane%8)*16, logical n = (cb_lin ^ ((row&3)<<5))/2 + n0
-  int d_row[2], d_n[2];
+  // D: 8 instrs: instr i covers rows i*8..i*8+7; per-wave share = DPI
+  int d_row[DPI], d_n[DPI];
   #pragma unroll
-  for (int j = 0; j < 2; ++j) {
-    int row = (w * 2 + j) * 8 + (lane >> 3);
+  for (int j = 0; j < DPI; ++j) {
+    int row = (w * DPI + j) * 8 + (lane >> 3);
     int cb = ((lane & 7) * 16) ^ DXOR(row);
     d_row[j] = row;
     d_n[j] = n0 + cb / 2;
@@ -1293,10 +1295,10 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
   // only at image borders / batch wraps (O(1/OH) of steps).
   const unsigned C2 = (unsigned)p.Cin * 2;
   const unsigned WC2 = (unsigned)p.W * C2;
-  int a_ow[4], a_oh[4], a_b[4], a_ih[4], a_iw[4];
-  unsigned avo[4];
-  bool ainb[4], avalid[4];
-  unsigned dvo[2];
+  int a_ow[API], a_oh[API], a_b[API], a_ih[API], a_iw[API];
+  unsigned avo[API];
+  bool ainb[API], avalid[API];
+  unsigned dvo[DPI];
 
   auto full_a = [&](int j) {
     int ih = a_ih[j], iw = a_iw[j];
@@ -1313,7 +1315,7 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
   };
 
   #pragma unroll
-  for (int j = 0; j < 4; ++j) {
+  for (int j = 0; j < API; ++j) {
     long m = mstart + a_row[j];
     a_ow[j] = (int)(m % p.OW);
     int t = (int)(m / p.OW);
@@ -1326,12 +1328,12 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
     full_a(j);
   }
   #pragma unroll
-  for (int j = 0; j < 2; ++j)
+  for (int j = 0; j < DPI; ++j)
     dvo[j] = (unsigned)(((mstart + d_row[j]) * p.Cout + d_n[j]) * 2);
 
   auto advance = [&]() {
     #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < API; ++j) {
       int ow0 = a_ow[j], oh0 = a_oh[j], b0 = a_b[j];
       a_ow[j] += WG_BM;
       while (a_ow[j] >= p.OW) { a_ow[j] -= p.OW; ++a_oh[j]; }
@@ -1350,31 +1352,31 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
       ainb[j] = inb;
     }
     #pragma unroll
-    for (int j = 0; j < 2; ++j) dvo[j] += (unsigned)(WG_BM * p.Cout * 2);
+    for (int j = 0; j < DPI; ++j) dvo[j] += (unsigned)(WG_BM * p.Cout * 2);
   };
 
   auto stage = [&](int buf, long ms) {
     #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < API; ++j) {
       unsigned vo = (a_kv[j] && avalid[j] && ms + a_row[j] < p.M)
                         ? avo[j] : 0xFF000000u;
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
-          rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * 4 + j) * 512],
+          rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * API + j) * 512],
           16, vo, 0, 0, 0);
     }
     #pragma unroll
-    for (int j = 0; j < 2; ++j) {
+    for (int j = 0; j < DPI; ++j) {
       unsigned vo = (d_n[j] < p.Cout && ms + d_row[j] < p.M)
                         ? dvo[j] : 0xFF000000u;
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
-          rd, (__attribute__((address_space(3))) void*)&sm.D[buf][(w * 2 + j) * 512],
+          rd, (__attribute__((address_space(3))) void*)&sm.D[buf][(w * DPI + j) * 512],
           16, vo, 0, 0, 0);
     }
   };
 
-  v4f acc[4][2] = {};
+  v4f acc[KF][2] = {};
   const int wr = w >> 1, wc = w & 1;
-  const int wk0 = wr * 64, wn0 = wc * 32;  // wave tile: 64 k x 32 n
+  const int wk0 = wr * (KF * 16), wn0 = wc * 32;
   const int fr = lane & 15, fg = lane >> 4;
   const int jg = lane & 15;
   // tr-read per-lane address components (bytes)
@@ -1395,9 +1397,9 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
     #pragma unroll
     for (int kk = 0; kk < WG_BM; kk += 32) {
       // A fragments: MFMA row = weight-k col0+fr, reduce elems = m
-      v8bf a[4];
+      v8bf a[KF];
       #pragma unroll
-      for (int kf = 0; kf < 4; ++kf) {
+      for (int kf = 0; kf < KF; ++kf) {
         int cbl = (wk0 + kf * 16) * 2 + tr_cb_a;
         int row = kk + fg * 8 + tr_row_a;
         v4bfx lo = tr16_read(Ab + row * 256 + (cbl ^ AXOR(row)));
@@ -1415,14 +1417,11 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
         v4bfx hi = tr16_read(Db + row * 128 + (cbl ^ DXOR(row)));
         bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[0], bfr[0], acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[0], bfr[1], acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[1], bfr[0], acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[1], bfr[1], acc[1][1], 0, 0, 0);
-      acc[2][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[2], bfr[0], acc[2][0], 0, 0, 0);
-      acc[2][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[2], bfr[1], acc[2][1], 0, 0, 0);
-      acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[3], bfr[0], acc[3][0], 0, 0, 0);
-      acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[3], bfr[1], acc[3][1], 0, 0, 0);
+      #pragma unroll
+      for (int kf = 0; kf < KF; ++kf) {
+        acc[kf][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[kf], bfr[0], acc[kf][0], 0, 0, 0);
+        acc[kf][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[kf], bfr[1], acc[kf][1], 0, 0, 0);
+      }
     }
     __syncthreads();
   }
@@ -1435,7 +1434,7 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_glds_kernel(WgradParams p) {
     for (int nf = 0; nf < 2; ++nf) {
       int nl = wn0 + nf * 16 + fr;
       #pragma unroll
-      for (int kf = 0; kf < 4; ++kf) {
+      for (int kf = 0; kf < KF; ++kf) {
         int kl = wk0 + kf * 16 + fg * 4;
         *(float4*)&p.ws[chunk + (long)nl * WG_BK + kl] =
             *(const float4*)&acc[kf][nf];
@@ -1771,7 +1770,10 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
                          WG_BN * WG_BK},
                         x.options().dtype(at::kFloat));
     p.ws = (float*)ws.mutable_data_ptr();
-    hipLaunchKernelGGL(wgrad_glds_kernel, grid, dim3(NTHREADS), 0, stream, p);
+    if (conv_nw() == 8)
+      hipLaunchKernelGGL((wgrad_glds_kernel<8>), grid, dim3(512), 0, stream, p);
+    else
+      hipLaunchKernelGGL((wgrad_glds_kernel<4>), grid, dim3(NTHREADS), 0, stream, p);
     long total = (long)p.ktiles * WG_BK * p.ntiles * WG_BN;
     hipLaunchKernelGGL(wgrad_reduce_kernel, dim3(cdiv(total, 256)),
                        dim3(256), 0, stream,

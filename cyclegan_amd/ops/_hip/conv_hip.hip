@@ -1573,16 +1573,16 @@ static void launch_glds_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
   if (use_pipe3()) {
     switch (p.stride) {
       case 1:
-        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 1, 3>), grid,
-                           dim3(NTHREADS), 0, stream, p);
+        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 1, 3, 8>), grid,
+                           dim3(512), 0, stream, p);
         return;
       case 2:
-        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 2, 3>), grid,
-                           dim3(NTHREADS), 0, stream, p);
+        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 2, 3, 8>), grid,
+                           dim3(512), 0, stream, p);
         return;
       default:
-        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 0, 3>), grid,
-                           dim3(NTHREADS), 0, stream, p);
+        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 0, 3, 8>), grid,
+                           dim3(512), 0, stream, p);
         return;
     }
   }

@@ -148,3 +148,32 @@ def test_conv_dtype_paths(dtype):
     w = torch.randn(4, 3, 3, 3)  # fp32 master
     y = ops.conv2d(x, w, padding="same")
     assert y.dtype == dtype
+
+
+def test_gradcheck_conv_small():
+    """fp64 gradcheck of the conv op (reference path) on tiny shapes."""
+    torch.manual_seed(8)
+    x = torch.randn(1, 5, 5, 2, dtype=torch.float64, requires_grad=True)
+    w = torch.randn(3, 3, 3, 2, dtype=torch.float64, requires_grad=True)
+    assert torch.autograd.gradcheck(
+        lambda a, b: ops.conv2d(a, b, None, 1, (1, 1, 1, 1), "reflect"),
+        (x, w), eps=1e-6, atol=1e-4)
+
+
+def test_gradcheck_conv_transpose_small():
+    torch.manual_seed(9)
+    x = torch.randn(1, 4, 4, 3, dtype=torch.float64, requires_grad=True)
+    w = torch.randn(2, 3, 3, 3, dtype=torch.float64, requires_grad=True)
+    assert torch.autograd.gradcheck(
+        lambda a, b: ops.conv_transpose2d(a, b, stride=2), (x, w),
+        eps=1e-6, atol=1e-4)
+
+
+def test_gradcheck_instance_norm_small():
+    torch.manual_seed(10)
+    x = torch.randn(2, 4, 4, 3, dtype=torch.float64, requires_grad=True)
+    g = torch.randn(3, dtype=torch.float64, requires_grad=True)
+    b = torch.randn(3, dtype=torch.float64, requires_grad=True)
+    assert torch.autograd.gradcheck(
+        lambda a, gg, bb: ops.instance_norm(a, gg, bb, eps=1e-3), (x, g, b),
+        eps=1e-6, atol=1e-4)

@@ -25,11 +25,14 @@ EPS_DEFAULT = 1e-3  # tfa InstanceNormalization default
 
 
 def _in_ref(x, gamma, beta, eps, act, slope, residual):
-    xf = x.float()
+    # compute in fp32 for bf16 inputs; keep native precision otherwise
+    # (fp64 gradcheck needs the graph to stay fp64)
+    cdt = torch.float32 if x.dtype == torch.bfloat16 else x.dtype
+    xf = x.to(cdt)
     mean = xf.mean(dim=(1, 2), keepdim=True)
     var = xf.var(dim=(1, 2), unbiased=False, keepdim=True)
     y = (xf - mean) * torch.rsqrt(var + eps)
-    y = y * gamma.float() + beta.float()
+    y = y * gamma.to(cdt) + beta.to(cdt)
     y = y.to(x.dtype)
     if residual is not None:
         y = y + residual

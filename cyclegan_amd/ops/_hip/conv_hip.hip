@@ -233,25 +233,33 @@ __global__ __launch_bounds__(NTHREADS) void conv_gemm_kernel(ConvParams p) {
 // the buffer descriptor turns into zeros. One __shared__ object only
 // (hipcc de-pipelines glds next to a second one).
 
-template <int NBUF>
+template <int NBUF, int BNT = BN>
 struct ConvSmemT {
-  short A[NBUF][BM * BK];
-  short Bt[NBUF][BN * BK];
+  // small arrays first: keeps their ds offsets within the 16-bit
+  // immediate range even when the tile images exceed 64 KiB
   long rowyb[BM];
   int rowih[BM], rowiw[BM];
   unsigned rowxb[BM];  // byte offset of batch base (tensors < 4 GiB)
   char rowok[BM];
+  char _pad[16 - (BM % 16 ? BM % 16 : 16)];
+  short A[NBUF][BM * BK];
+  short Bt[NBUF][BNT * BK];
 };
 using ConvSmem = ConvSmemT<2>;
 
-template <bool IS_CONVT, int STRIDE, int NBUF = 2, int NW = 4>
+template <bool IS_CONVT, int STRIDE, int NBUF = 2, int NW = 4, int BNT = BN>
 __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
   constexpr int NT = NW * 64;        // threads
   constexpr int API = 16 / NW;       // A glds instructions per wave
-  constexpr int BPI = 8 / NW;        // B glds instructions per wave
-  constexpr int MF = (256 / NW) / 16;  // m-fragments per wave
+  constexpr int BPI = (BNT / 8) / NW;  // B glds instructions per wave
+  constexpr int NWC = BNT / 32;      // wave-grid columns (32-n wave tiles)
+  constexpr int NWR = NW / NWC;      // wave-grid rows
+  constexpr int MF = (BM / NWR) / 16;  // m-fragments per wave
   const int stride = STRIDE ? STRIDE : p.stride;
-  __shared__ ConvSmemT<NBUF> sm;
+  // dynamic LDS (the BNT=128 image exceeds the 64 KiB static limit);
+  // ONE region, 16-B aligned (G17), opted-in at launch time.
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  auto& sm = *reinterpret_cast<ConvSmemT<NBUF, BNT>*>(smem_raw);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -259,7 +267,7 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
   const int bid = blockIdx.x;
   const int mt = bid % p.mtiles, nt = bid / p.mtiles;
   const long m0 = (long)mt * BM;
-  const int n0 = nt * BN;
+  const int n0 = nt * BNT;
 
   for (int r = tid; r < BM; r += NT) {
     long m = m0 + r;
@@ -298,6 +306,7 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
   int bn[BPI];
   #pragma unroll
   for (int j = 0; j < BPI; ++j) bn[j] = (w * BPI + j) * 8 + lr + n0;
+  static_assert(BNT == 64 || BNT == 128, "BNT");
 
   auto rx = __builtin_amdgcn_make_buffer_rsrc(
       (void*)p.x, 0, (unsigned)((long)p.B * p.H * p.W * p.Cin * 2), 0x00020000);
@@ -403,7 +412,7 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
   };
 
   v4f acc[MF][2] = {};
-  const int wr = w >> 1, wc = w & 1;
+  const int wr = w / NWC, wc = w % NWC;
   const int wm0 = wr * (MF * 16), wn0 = wc * 32;
   const int fr = lane & 15;
   const int fg = lane >> 4;
@@ -452,6 +461,13 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
       for (int r = 0; r < 4; ++r) {
         int rl = wm0 + mf * 16 + fg * 4 + r;
         if (!sm.rowok[rl]) continue;
+#ifdef CYG_DEBUG_EPI
+        if (sm.rowyb[rl] < 0 || sm.rowyb[rl] + n >= p.M * p.Cout) {
+          printf("EPI OOB bid=%d tid=%d rl=%d rowyb=%ld n=%d\n",
+                 bid, tid, rl, (long)sm.rowyb[rl], n);
+          continue;
+        }
+#endif
         float v = apply_act(acc[mf][nf][r] + bv, p.act, p.slope);
         p.y[sm.rowyb[rl] + n] = f2b(v);
       }
@@ -1552,52 +1568,60 @@ static int conv_nw() {
   return v;
 }
 
+static int conv_bn() {
+  static int v = []() {
+    const char* e = getenv("CYG_CONV_BN");
+    return e ? atoi(e) : 128;
+  }();
+  return v;
+}
+
+template <bool IS_CONVT, int STRIDE, int NBUF, int NW, int BNT>
+static void launch_one_glds(const ConvParams& p, dim3 grid,
+                            hipStream_t stream) {
+  constexpr size_t SMB = sizeof(ConvSmemT<NBUF, BNT>);
+  static bool init = []() {
+    hipFuncSetAttribute(
+        (const void*)(conv_glds_kernel<IS_CONVT, STRIDE, NBUF, NW, BNT>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)SMB);
+    return true;
+  }();
+  (void)init;
+  hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, STRIDE, NBUF, NW, BNT>),
+                     grid, dim3(NW * 64), SMB, stream, p);
+}
+
 template <bool IS_CONVT>
 static void launch_glds_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
+  if (conv_nw() == 8 && conv_bn() == 128 && (p.Cout % 128) == 0) {
+    // wider n-tile: 1.5x arithmetic intensity for Cout >= 128 layers
+    ConvParams q = p;
+    q.ntiles = (p.Cout + 127) / 128;
+    dim3 g2((long)q.mtiles * q.ntiles);
+    switch (p.stride) {
+      case 1: launch_one_glds<IS_CONVT, 1, 2, 8, 128>(q, g2, stream); return;
+      case 2: launch_one_glds<IS_CONVT, 2, 2, 8, 128>(q, g2, stream); return;
+      default: launch_one_glds<IS_CONVT, 0, 2, 8, 128>(q, g2, stream); return;
+    }
+  }
   if (conv_nw() == 8) {
     switch (p.stride) {
-      case 1:
-        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 1, 2, 8>), grid,
-                           dim3(512), 0, stream, p);
-        return;
-      case 2:
-        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 2, 2, 8>), grid,
-                           dim3(512), 0, stream, p);
-        return;
-      default:
-        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 0, 2, 8>), grid,
-                           dim3(512), 0, stream, p);
-        return;
+      case 1: launch_one_glds<IS_CONVT, 1, 2, 8, 64>(p, grid, stream); return;
+      case 2: launch_one_glds<IS_CONVT, 2, 2, 8, 64>(p, grid, stream); return;
+      default: launch_one_glds<IS_CONVT, 0, 2, 8, 64>(p, grid, stream); return;
     }
   }
   if (use_pipe3()) {
     switch (p.stride) {
-      case 1:
-        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 1, 3, 8>), grid,
-                           dim3(512), 0, stream, p);
-        return;
-      case 2:
-        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 2, 3, 8>), grid,
-                           dim3(512), 0, stream, p);
-        return;
-      default:
-        hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 0, 3, 8>), grid,
-                           dim3(512), 0, stream, p);
-        return;
+      case 1: launch_one_glds<IS_CONVT, 1, 3, 8, 64>(p, grid, stream); return;
+      case 2: launch_one_glds<IS_CONVT, 2, 3, 8, 64>(p, grid, stream); return;
+      default: launch_one_glds<IS_CONVT, 0, 3, 8, 64>(p, grid, stream); return;
     }
   }
   switch (p.stride) {
-    case 1:
-      hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 1>), grid,
-                         dim3(NTHREADS), 0, stream, p);
-      break;
-    case 2:
-      hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 2>), grid,
-                         dim3(NTHREADS), 0, stream, p);
-      break;
-    default:
-      hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, 0>), grid,
-                         dim3(NTHREADS), 0, stream, p);
+    case 1: launch_one_glds<IS_CONVT, 1, 2, 4, 64>(p, grid, stream); break;
+    case 2: launch_one_glds<IS_CONVT, 2, 2, 4, 64>(p, grid, stream); break;
+    default: launch_one_glds<IS_CONVT, 0, 2, 4, 64>(p, grid, stream);
   }
 }
 

@@ -26,7 +26,8 @@ if sources:
         sources=sources,
         extra_compile_args={
             "cxx": ["-O3", "-std=c++17"],
-            "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+            "nvcc": (["-O3", "-std=c++17", "--offload-arch=gfx950"] +
+                     os.environ.get("CYG_EXTRA_FLAGS", "").split()),
         },
     ))
 

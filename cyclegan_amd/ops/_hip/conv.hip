@@ -1248,17 +1248,22 @@ DEV v4bfx tr16_read(const char* plds) {
       (__attribute__((address_space(3))) v4bfx*)plds);
 }
 
-struct WgSmem {
+template <int WBN>
+struct WgSmemT {
   short A[2][WG_BM * WG_BK];   // [m][k] swizzled
-  short D[2][WG_BM * WG_BN];   // [m][n] swizzled
+  short D[2][WG_BM * WBN];     // [m][n] swizzled
 };
 
-template <int NW = 4>
+template <int NW = 4, int WBN = 64>
 __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
-  constexpr int API = 16 / NW;   // A glds instructions per wave
-  constexpr int DPI = 8 / NW;    // D glds instructions per wave
-  constexpr int KF = (256 / NW) / 16;  // k-fragments per wave
-  __shared__ WgSmem sm;
+  constexpr int API = 16 / NW;        // A glds instructions per wave
+  constexpr int DCH = WBN / 8;        // 16-B chunks per D row
+  constexpr int DRPI = 64 / DCH;      // D rows per glds instruction
+  constexpr int DPI = (WG_BM / DRPI) / NW;  // D instructions per wave
+  constexpr int NWCW = WBN / 32;      // wave-grid columns
+  constexpr int KF = (WG_BK / (NW / NWCW)) / 16;  // k-fragments per wave
+  extern __shared__ __attribute__((aligned(16))) char wg_smem_raw[];
+  auto& sm = *reinterpret_cast<WgSmemT<WBN>*>(wg_smem_raw);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -1267,7 +1272,7 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   const int nt = (blockIdx.x / p.ktiles) % p.ntiles;
   const int sl = blockIdx.x / (p.ktiles * p.ntiles);
   const long k0 = (long)kt * WG_BK;
-  const int n0 = nt * WG_BN;
+  const int n0 = nt * WBN;
   const long mstart = sl * p.mchunks_per_slice * WG_BM;
   long mend = mstart + p.mchunks_per_slice * WG_BM;
   if (mend > p.M) mend = p.M;
@@ -1295,12 +1300,12 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     a_dkw[j] = tap - a_dkh[j] * p.KW;
     a_k[j] = (int)(k - k0);
   }
-  // D: 8 instrs: instr i covers rows i*8..i*8+7; per-wave share = DPI
+  // D: instr i covers DRPI rows; per-wave share = DPI
   int d_row[DPI], d_n[DPI];
   #pragma unroll
   for (int j = 0; j < DPI; ++j) {
-    int row = (w * DPI + j) * 8 + (lane >> 3);
-    int cb = ((lane & 7) * 16) ^ DXOR(row);
+    int row = (w * DPI + j) * DRPI + lane / DCH;
+    int cb = ((lane % DCH) * 16) ^ (WBN == 128 ? AXOR(row) : DXOR(row));
     d_row[j] = row;
     d_n[j] = n0 + cb / 2;
   }
@@ -1390,7 +1395,7 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   };
 
   v4f acc[KF][2] = {};
-  const int wr = w >> 1, wc = w & 1;
+  const int wr = w / NWCW, wc = w % NWCW;
   const int wk0 = wr * (KF * 16), wn0 = wc * 32;
   const int fr = lane & 15, fg = lane >> 4;
   const int jg = lane & 15;
@@ -1427,9 +1432,11 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
       for (int nf = 0; nf < 2; ++nf) {
         int cbl = (wn0 + nf * 16) * 2 + tr_cb_a;
         int row = kk + fg * 8 + tr_row_a;
-        v4bfx lo = tr16_read(Db + row * 128 + (cbl ^ DXOR(row)));
+        v4bfx lo = tr16_read(Db + row * (WBN * 2) +
+                             (cbl ^ (WBN == 128 ? AXOR(row) : DXOR(row))));
         row += 4;
-        v4bfx hi = tr16_read(Db + row * 128 + (cbl ^ DXOR(row)));
+        v4bfx hi = tr16_read(Db + row * (WBN * 2) +
+                             (cbl ^ (WBN == 128 ? AXOR(row) : DXOR(row))));
         bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
       #pragma unroll
@@ -1441,10 +1448,10 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     __syncthreads();
   }
 
-  // slab store (no atomics): chunk layout [WG_BN][WG_BK], float4 rows
+  // slab store (no atomics): chunk layout [WBN][WG_BK], float4 rows
   {
     long chunk = (((long)sl * p.ktiles + kt) * p.ntiles + nt) *
-                 ((long)WG_BN * WG_BK);
+                 ((long)WBN * WG_BK);
     #pragma unroll
     for (int nf = 0; nf < 2; ++nf) {
       int nl = wn0 + nf * 16 + fr;
@@ -1462,22 +1469,22 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
 __global__ void wgrad_reduce_kernel(const float* __restrict__ ws,
                                     float* __restrict__ dw, long KTOT,
                                     int Cout, int ktiles, int ntiles,
-                                    int slices) {
+                                    int slices, int wbn) {
   long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long total = (long)ktiles * WG_BK * ntiles * WG_BN;
+  long total = (long)ktiles * WG_BK * ntiles * wbn;
   if (idx >= total) return;
   // idx -> (nt, nl, kt, kl) with kl fastest for coalescing
   int kl = (int)(idx % WG_BK);
   long t = idx / WG_BK;
   int kt = (int)(t % ktiles);
   t /= ktiles;
-  int nl = (int)(t % WG_BN);
-  int nt = (int)(t / WG_BN);
+  int nl = (int)(t % wbn);
+  int nt = (int)(t / wbn);
   long k = (long)kt * WG_BK + kl;
-  int n = nt * WG_BN + nl;
+  int n = nt * wbn + nl;
   if (k >= KTOT || n >= Cout) return;
-  long stride = (long)ktiles * ntiles * WG_BN * WG_BK;
-  long off = (((long)kt * ntiles + nt) * WG_BN + nl) * WG_BK + kl;
+  long stride = (long)ktiles * ntiles * wbn * WG_BK;
+  long off = (((long)kt * ntiles + nt) * wbn + nl) * WG_BK + kl;
   float s = 0;
   for (int sl = 0; sl < slices; ++sl) s += ws[off + sl * stride];
   dw[(long)n * KTOT + k] = s;
@@ -1789,19 +1796,48 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
     auto dw = at::empty({p.Cout, (long)KH, (long)KW, p.Cin},
                         x.options().dtype(at::kFloat));
     p.dw = (float*)dw.mutable_data_ptr();
+    int wbn = (conv_nw() == 8 && conv_bn() == 128 && (p.Cout % 128) == 0)
+                  ? 128 : WG_BN;
+    if (wbn == 128) {
+      p.ntiles = (p.Cout + 127) / 128;
+      // re-balance the split-M slices for the halved tile count
+      long mchunks = (p.M + WG_BM - 1) / WG_BM;
+      static int tgt2 = []() {
+        const char* e = getenv("CYG_WG_BLOCKS");
+        return e ? atoi(e) : 512;
+      }();
+      int target = std::max<long>(1, tgt2 / ((long)p.ktiles * p.ntiles));
+      p.slices = (int)std::min<long>(mchunks, target);
+      p.mchunks_per_slice = (mchunks + p.slices - 1) / p.slices;
+      grid = dim3((long)p.ktiles * p.ntiles * p.slices);
+    }
     auto ws = at::empty({(long)p.slices * p.ktiles * p.ntiles *
-                         WG_BN * WG_BK},
+                         wbn * WG_BK},
                         x.options().dtype(at::kFloat));
     p.ws = (float*)ws.mutable_data_ptr();
-    if (conv_nw() == 8)
-      hipLaunchKernelGGL((wgrad_glds_kernel<8>), grid, dim3(512), 0, stream, p);
-    else
-      hipLaunchKernelGGL((wgrad_glds_kernel<4>), grid, dim3(NTHREADS), 0, stream, p);
-    long total = (long)p.ktiles * WG_BK * p.ntiles * WG_BN;
+    if (wbn == 128) {
+      constexpr size_t SMB = sizeof(WgSmemT<128>);
+      static bool init = []() {
+        hipFuncSetAttribute((const void*)(wgrad_glds_kernel<8, 128>),
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            (int)SMB);
+        return true;
+      }();
+      (void)init;
+      hipLaunchKernelGGL((wgrad_glds_kernel<8, 128>), grid, dim3(512), SMB,
+                         stream, p);
+    } else if (conv_nw() == 8) {
+      hipLaunchKernelGGL((wgrad_glds_kernel<8, 64>), grid, dim3(512),
+                         sizeof(WgSmemT<64>), stream, p);
+    } else {
+      hipLaunchKernelGGL((wgrad_glds_kernel<4, 64>), grid, dim3(NTHREADS),
+                         sizeof(WgSmemT<64>), stream, p);
+    }
+    long total = (long)p.ktiles * WG_BK * p.ntiles * wbn;
     hipLaunchKernelGGL(wgrad_reduce_kernel, dim3(cdiv(total, 256)),
                        dim3(256), 0, stream,
                        (const float*)ws.const_data_ptr(), p.dw, p.KTOT,
-                       p.Cout, p.ktiles, p.ntiles, p.slices);
+                       p.Cout, p.ktiles, p.ntiles, p.slices, wbn);
     return dw;
   }
   auto dw = at::zeros({p.Cout, (long)KH, (long)KW, p.Cin},

@@ -214,7 +214,8 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
     const short* __restrict__ yact, const float* __restrict__ gamma,
     const float* __restrict__ mean,
     const float* __restrict__ rstd, const float* __restrict__ p1,
-    const float* __restrict__ p2, short* __restrict__ dx, int B, long HW,
+    const float* __restrict__ p2, short* __restrict__ dx,
+    float* __restrict__ dbeta, float* __restrict__ dgamma, int B, long HW,
     int C, int S, int act, float slope) {
   int b = blockIdx.x / S;
   int sl = blockIdx.x % S;
@@ -249,6 +250,10 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
     sm2[c] = t2 * inv_hw;
     smean[c] = mean[(long)b * C + c];
     srstd[c] = rstd[(long)b * C + c];
+    if (sl == 0) {  // fold the dgamma/dbeta reduction in (B adders per c)
+      atomicAdd(&dbeta[c], t1);
+      atomicAdd(&dgamma[c], t2);
+    }
   }
   __syncthreads();
 
@@ -577,14 +582,10 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
                      (const float*)rstd.const_data_ptr(),
                      (const float*)p1.const_data_ptr(),
                      (const float*)p2.const_data_ptr(),
-                     (short*)dx.mutable_data_ptr(), B, HW, C, S, (int)act,
-                     (float)slope);
-  hipLaunchKernelGGL(in_bwd_dgb_kernel,
-                     dim3(cdiv64(C, 256), cdiv64(S * B, 8)), dim3(256), 0,
-                     stream, (const float*)p1.const_data_ptr(),
-                     (const float*)p2.const_data_ptr(),
+                     (short*)dx.mutable_data_ptr(),
                      (float*)dbeta.mutable_data_ptr(),
-                     (float*)dgamma.mutable_data_ptr(), S * B, C);
+                     (float*)dgamma.mutable_data_ptr(), B, HW, C, S,
+                     (int)act, (float)slope);
   return {dx, dgamma, dbeta};
 }
 

@@ -246,7 +246,8 @@ struct ConvSmemT {
 };
 using ConvSmem = ConvSmemT<2>;
 
-template <bool IS_CONVT, int STRIDE, int NBUF = 2, int NW = 4, int BNT = BN>
+template <bool IS_CONVT, int STRIDE, int NBUF = 2, int NW = 4, int BNT = BN,
+          bool PHASED = false>
 __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
   constexpr int NT = NW * 64;        // threads
   constexpr int API = 16 / NW;       // A glds instructions per wave
@@ -265,25 +266,54 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
   const int w = tid >> 6;
   const int bid = blockIdx.x;
   const int mt = bid % p.mtiles, nt = bid / p.mtiles;
-  const long m0 = (long)mt * BM;
   const int n0 = nt * BNT;
 
+  // PHASED (convT stride-2 only): the M space is regrouped into 4 output
+  // phase classes; each tile enumerates ONLY its phase's taps (KEFF =
+  // nvh*nvw*Cin), removing the 3/4 structural-zero gather waste.
+  static_assert(!PHASED || (IS_CONVT && STRIDE == 2), "PHASED");
+  const int tpp = PHASED ? (p.mtiles >> 2) : 0;
+  const int phase = PHASED ? mt / tpp : 0;
+  const int phh = phase >> 1, phw = phase & 1;
+  const int OH2 = p.OH >> 1, OW2 = p.OW >> 1;
+  const long m0 = PHASED ? (long)(mt - phase * tpp) * BM : (long)mt * BM;
+  const int nvh = PHASED ? ((p.KH - phh + 1) >> 1) : 0;
+  const int nvw = PHASED ? ((p.KW - phw + 1) >> 1) : 0;
+  const long KEFF = PHASED ? (long)nvh * nvw * p.Cin : p.KTOT;
+
   for (int r = tid; r < BM; r += NT) {
-    long m = m0 + r;
-    bool ok = m < p.M;
-    long mm = ok ? m : 0;
-    int ow = (int)(mm % p.OW);
-    int oh = (int)((mm / p.OW) % p.OH);
-    int b = (int)(mm / ((long)p.OW * p.OH));
-    sm.rowok[r] = ok;
-    sm.rowxb[r] = (unsigned)((long)b * p.H * p.W * p.Cin * 2);
-    sm.rowyb[r] = ((long)(b * p.OH + oh) * p.OW + ow) * p.Cout;
-    if (IS_CONVT) {
-      sm.rowih[r] = oh + p.pt;
-      sm.rowiw[r] = ow + p.pl;
+    if (PHASED) {
+      long mp = m0 + r;
+      long Mp = (long)p.B * OH2 * OW2;
+      bool ok = mp < Mp;
+      long mm = ok ? mp : 0;
+      int owp = (int)(mm % OW2);
+      int ohp = (int)((mm / OW2) % OH2);
+      int b = (int)(mm / ((long)OW2 * OH2));
+      int i = 2 * ohp + ((phh + p.pt) & 1);
+      int j = 2 * owp + ((phw + p.pl) & 1);
+      sm.rowok[r] = ok && i < p.OH && j < p.OW;
+      sm.rowxb[r] = (unsigned)((long)b * p.H * p.W * p.Cin * 2);
+      sm.rowyb[r] = ((long)(b * p.OH + i) * p.OW + j) * p.Cout;
+      sm.rowih[r] = (i + p.pt - phh) >> 1;  // oh = this - vh
+      sm.rowiw[r] = (j + p.pl - phw) >> 1;
     } else {
-      sm.rowih[r] = oh * stride - p.pt;
-      sm.rowiw[r] = ow * stride - p.pl;
+      long m = m0 + r;
+      bool ok = m < p.M;
+      long mm = ok ? m : 0;
+      int ow = (int)(mm % p.OW);
+      int oh = (int)((mm / p.OW) % p.OH);
+      int b = (int)(mm / ((long)p.OW * p.OH));
+      sm.rowok[r] = ok;
+      sm.rowxb[r] = (unsigned)((long)b * p.H * p.W * p.Cin * 2);
+      sm.rowyb[r] = ((long)(b * p.OH + oh) * p.OW + ow) * p.Cout;
+      if (IS_CONVT) {
+        sm.rowih[r] = oh + p.pt;
+        sm.rowiw[r] = ow + p.pl;
+      } else {
+        sm.rowih[r] = oh * stride - p.pt;
+        sm.rowiw[r] = ow * stride - p.pl;
+      }
     }
   }
   __syncthreads();
@@ -312,14 +342,14 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
   auto rw = __builtin_amdgcn_make_buffer_rsrc(
       (void*)p.w, 0, (unsigned)(p.Cout * p.KTOT * 2), 0x00020000);
 
-  const int nk = (int)((p.KTOT + BK - 1) / BK);
+  const int nk = (int)((KEFF + BK - 1) / BK);
   const bool big_ci = p.Cin >= BK;  // wave-uniform
 
   // incremental gather state: recompute voffsets only at tap boundaries
   // (every Cin/64 K-steps); otherwise one predicated add per instruction.
   long kcur;
   bool kv;
-  int ci, dkh, dkw;
+  int ci, dkh, dkw;  // PHASED: dkh/dkw hold (vh, vw)
   unsigned avo[API];
   bool avalid[API];
   unsigned bvo[BPI];
@@ -333,7 +363,10 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
       if (!aok[j]) continue;
       bool valid = true;
       int ih, iw;
-      if (IS_CONVT) {
+      if (PHASED) {
+        ih = aih[j] - dkh; iw = aiw[j] - dkw;  // oh = base - vh
+        valid = ih >= 0 && iw >= 0 && ih < p.H && iw < p.W;
+      } else if (IS_CONVT) {
         int nh = aih[j] - dkh, nw = aiw[j] - dkw;
         valid = nh >= 0 && nw >= 0 && (nh % stride) == 0 && (nw % stride) == 0;
         ih = nh / stride; iw = nw / stride;
@@ -353,43 +386,62 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
     }
   };
 
-  auto init_state = [&]() {
-    kcur = klog;
-    kv = kcur < p.KTOT;
+  auto recompute_b = [&]() {
+    long kw_idx = kcur;
+    if (PHASED)
+      kw_idx = ((long)(2 * dkh + phh) * p.KW + (2 * dkw + phw)) * p.Cin + ci;
+    #pragma unroll
+    for (int j = 0; j < BPI; ++j)
+      bvo[j] = (unsigned)(((long)bn[j] * p.KTOT + kw_idx) * 2);
+  };
+
+  auto decode_tap = [&]() {
     int tap = (int)(kcur / p.Cin);
     ci = (int)(kcur - (long)tap * p.Cin);
-    dkh = tap / p.KW;
-    dkw = tap - dkh * p.KW;
+    if (PHASED) {
+      dkh = tap / nvw;          // vh
+      dkw = tap - dkh * nvw;    // vw
+    } else {
+      dkh = tap / p.KW;
+      dkw = tap - dkh * p.KW;
+    }
+  };
+
+  auto init_state = [&]() {
+    kcur = klog;
+    kv = kcur < KEFF;
+    decode_tap();
     recompute_a();
     #pragma unroll
-    for (int j = 0; j < BPI; ++j) {
-      bnv[j] = bn[j] < p.Cout;
-      bvo[j] = (unsigned)(((long)bn[j] * p.KTOT + kcur) * 2);
-    }
+    for (int j = 0; j < BPI; ++j) bnv[j] = bn[j] < p.Cout;
+    recompute_b();
   };
 
   auto advance = [&]() {
     kcur += BK;
-    kv = kcur < p.KTOT;
-    #pragma unroll
-    for (int j = 0; j < BPI; ++j) bvo[j] += BK * 2;
+    kv = kcur < KEFF;
     if (big_ci) {
       ci += BK;
       if (ci >= p.Cin) {
         ci -= p.Cin;
-        if (++dkw == p.KW) { dkw = 0; ++dkh; }
+        if (PHASED) {
+          if (++dkw == nvw) { dkw = 0; ++dkh; }
+        } else {
+          if (++dkw == p.KW) { dkw = 0; ++dkh; }
+        }
         recompute_a();
+        recompute_b();
       } else {
         #pragma unroll
         for (int j = 0; j < API; ++j) avo[j] += BK * 2;
+        #pragma unroll
+        for (int j = 0; j < BPI; ++j) bvo[j] += BK * 2;
       }
     } else {
       // Cin < 64: a K-step crosses several taps — full recompute
-      int tap = (int)(kcur / p.Cin);
-      ci = (int)(kcur - (long)tap * p.Cin);
-      dkh = tap / p.KW;
-      dkw = tap - dkh * p.KW;
+      decode_tap();
       recompute_a();
+      recompute_b();
     }
   };
 
@@ -704,234 +756,6 @@ __global__ __launch_bounds__(NTHREADS) void conv_fp8_kernel(ConvParams p) {
         int rl = wm0 + mf * 16 + fg * 4 + r;
         if (!sm.rowok[rl]) continue;
         float v = apply_act(acc[mf][nf][r] * dqs + bv, p.act, p.slope);
-        p.y[sm.rowyb[rl] + n] = f2b(v);
-      }
-    }
-  }
-}
-
-// ---------------- phase-decomposed stride-2 transpose conv ----------------
-// convT s2 (and s2 conv dgrad) gathers hit only taps whose parity matches
-// the output pixel's phase — 3/4 of a dense gather is zeros. Here the M
-// space is regrouped into 4 phase classes (p.mtiles = 4 * tiles/phase, OH
-// and OW even), each tile enumerates ONLY its phase's taps
-// (K_eff = nvh*nvw*Cin), so every MFMA operates on useful data.
-__global__ __launch_bounds__(NTHREADS) void convt_phased_kernel(ConvParams p) {
-  constexpr int NBUF = 2;  // (shares the templated loop body text)
-  constexpr int MF = 4;
-  __shared__ ConvSmem sm;
-
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int w = tid >> 6;
-  const int bid = blockIdx.x;
-  const int mt = bid % p.mtiles, nt = bid / p.mtiles;
-  const int tpp = p.mtiles >> 2;          // tiles per phase
-  const int phase = mt / tpp;
-  const int ph = phase >> 1, pw = phase & 1;
-  const int OH2 = p.OH >> 1, OW2 = p.OW >> 1;
-  const long Mp = (long)p.B * OH2 * OW2;  // pixels per phase
-  const long m0 = (long)(mt - phase * tpp) * BM;
-  const int n0 = nt * BN;
-  // valid tap counts for this phase: dk = 2*v + parity, parity = (ph+pt)&1
-  const int par_h = (ph + p.pt) & 1 ? ((p.pt + 1) & 1) : 0;  // see below
-  // dkh must satisfy (i + pt - dkh) even where (i+pt)%2 == phh_sel;
-  // phases are indexed directly by (i+pt)&1, so dkh parity == phase bit.
-  const int dkh_par = ph, dkw_par = pw;
-  const int nvh = (p.KH - dkh_par + 1) >> 1;
-  const int nvw = (p.KW - dkw_par + 1) >> 1;
-  const long KEFF = (long)nvh * nvw * p.Cin;
-  (void)par_h;
-
-  for (int r = tid; r < BM; r += NTHREADS) {
-    long mp = m0 + r;
-    bool ok = mp < Mp;
-    long mm = ok ? mp : 0;
-    int owp = (int)(mm % OW2);
-    int ohp = (int)((mm / OW2) % OH2);
-    int b = (int)(mm / ((long)OW2 * OH2));
-    // output coords of this phase: (i+pt)&1 == ph, (j+pl)&1 == pw
-    int i = 2 * ohp + ((ph + p.pt) & 1);
-    int j = 2 * owp + ((pw + p.pl) & 1);
-    sm.rowok[r] = ok && i < p.OH && j < p.OW;
-    sm.rowxb[r] = (unsigned)((long)b * p.H * p.W * p.Cin * 2);
-    sm.rowyb[r] = ((long)(b * p.OH + i) * p.OW + j) * p.Cout;
-    sm.rowih[r] = (i + p.pt - dkh_par) >> 1;  // oh = this - vh
-    sm.rowiw[r] = (j + p.pl - dkw_par) >> 1;
-  }
-  __syncthreads();
-
-  const int lr = lane >> 3;
-  const int klog = ((lane & 7) ^ lr) << 3;
-  int aih[4], aiw[4];
-  unsigned axb[4];
-  bool aok[4];
-  #pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    int r = w * 32 + j * 8 + lr;
-    aih[j] = sm.rowih[r];
-    aiw[j] = sm.rowiw[r];
-    axb[j] = sm.rowxb[r];
-    aok[j] = sm.rowok[r];
-  }
-  const int bn[2] = {(w * 2 + 0) * 8 + lr + n0, (w * 2 + 1) * 8 + lr + n0};
-
-  auto rx = __builtin_amdgcn_make_buffer_rsrc(
-      (void*)p.x, 0, (unsigned)((long)p.B * p.H * p.W * p.Cin * 2), 0x00020000);
-  auto rw = __builtin_amdgcn_make_buffer_rsrc(
-      (void*)p.w, 0, (unsigned)(p.Cout * p.KTOT * 2), 0x00020000);
-
-  const int nk = (int)((KEFF + BK - 1) / BK);
-  const bool big_ci = p.Cin >= BK;
-
-  long kcur;
-  bool kv;
-  int ci, vh, vw;
-  unsigned avo[4], bvo[2];
-  bool avalid[4], bnv[2];
-
-  auto recompute = [&]() {
-    int oh_off = vh, ow_off = vw;
-    #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      avalid[j] = false;
-      avo[j] = 0xFF000000u;
-      if (!aok[j]) continue;
-      int oh = aih[j] - oh_off, ow = aiw[j] - ow_off;
-      if (oh >= 0 && oh < p.H && ow >= 0 && ow < p.W) {
-        avalid[j] = true;
-        avo[j] = axb[j] + (unsigned)((((long)oh * p.W + ow) * p.Cin + ci) * 2);
-      }
-    }
-    // weight k index for this tap
-    long kw_idx = ((long)(2 * vh + dkh_par) * p.KW + (2 * vw + dkw_par)) * p.Cin + ci;
-    #pragma unroll
-    for (int j = 0; j < 2; ++j)
-      bvo[j] = (unsigned)(((long)bn[j] * p.KTOT + kw_idx) * 2);
-  };
-
-  auto init_state = [&]() {
-    kcur = klog;
-    kv = kcur < KEFF;
-    int vtap = (int)(kcur / p.Cin);
-    ci = (int)(kcur - (long)vtap * p.Cin);
-    vh = vtap / nvw;
-    vw = vtap - vh * nvw;
-    #pragma unroll
-    for (int j = 0; j < 2; ++j) bnv[j] = bn[j] < p.Cout;
-    recompute();
-  };
-
-  auto advance = [&]() {
-    kcur += BK;
-    kv = kcur < KEFF;
-    if (big_ci) {
-      ci += BK;
-      if (ci >= p.Cin) {
-        ci -= p.Cin;
-        if (++vw == nvw) { vw = 0; ++vh; }
-        recompute();
-      } else {
-        #pragma unroll
-        for (int j = 0; j < 4; ++j) avo[j] += BK * 2;
-        #pragma unroll
-        for (int j = 0; j < 2; ++j) bvo[j] += BK * 2;
-      }
-    } else {
-      int vtap = (int)(kcur / p.Cin);
-      ci = (int)(kcur - (long)vtap * p.Cin);
-      vh = vtap / nvw;
-      vw = vtap - vh * nvw;
-      recompute();
-    }
-  };
-
-  auto stage = [&](int buf) {
-    #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      unsigned vo = (kv && avalid[j]) ? avo[j] : 0xFF000000u;
-      __builtin_amdgcn_raw_ptr_buffer_load_lds(
-          rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * 4 + j) * 512],
-          16, vo, 0, 0, 0);
-    }
-    #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      unsigned vo = (kv && bnv[j]) ? bvo[j] : 0xFF000000u;
-      __builtin_amdgcn_raw_ptr_buffer_load_lds(
-          rw, (__attribute__((address_space(3))) void*)&sm.Bt[buf][(w * 2 + j) * 512],
-          16, vo, 0, 0, 0);
-    }
-  };
-
-  v4f acc[4][2] = {};
-  const int wr = w >> 1, wc = w & 1;
-  const int wm0 = wr * 64, wn0 = wc * 32;
-  const int fr = lane & 15;
-  const int fg = lane >> 4;
-  const int swz = (fr & 7) << 4;
-
-  init_state();
-  if constexpr (NBUF == 2) {
-    stage(0);
-    __syncthreads();
-  } else {
-    // 3-buffer: keep one tile's LDS-DMA in flight ACROSS the barrier
-    // (counted vmcnt + raw s_barrier; __syncthreads would drain it)
-    stage(0);
-    if (nk > 1) { advance(); stage(1); }
-  }
-
-  for (int kt = 0; kt < nk; ++kt) {
-    if constexpr (NBUF == 3) {
-      if (kt + 1 < nk)
-        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");  // tile kt landed
-      else
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      __builtin_amdgcn_s_barrier();
-      if (kt + 2 < nk) {
-        advance();
-        stage((kt + 2) % 3);
-      }
-    } else {
-      if (kt + 1 < nk) {
-        advance();
-        stage((kt + 1) & 1);
-      }
-    }
-    const char* Ab = (const char*)sm.A[NBUF == 3 ? kt % 3 : (kt & 1)];
-    const char* Bb = (const char*)sm.Bt[NBUF == 3 ? kt % 3 : (kt & 1)];
-    #pragma unroll
-    for (int kk = 0; kk < BK; kk += 32) {
-      const int kbyte = (kk + fg * 8) * 2;
-      v8bf a[MF], b[2];
-      #pragma unroll
-      for (int mf = 0; mf < MF; ++mf)
-        a[mf] = *(const v8bf*)(Ab + ((wm0 + mf * 16 + fr) << 7) + (kbyte ^ swz));
-      #pragma unroll
-      for (int nf = 0; nf < 2; ++nf)
-        b[nf] = *(const v8bf*)(Bb + ((wn0 + nf * 16 + fr) << 7) + (kbyte ^ swz));
-      #pragma unroll
-      for (int mf = 0; mf < MF; ++mf) {
-        acc[mf][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mf], b[0], acc[mf][0], 0, 0, 0);
-        acc[mf][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mf], b[1], acc[mf][1], 0, 0, 0);
-      }
-    }
-    if constexpr (NBUF == 2) __syncthreads();
-  }
-  if constexpr (NBUF == 3) __syncthreads();  // before epilogue rowyb reads
-
-  #pragma unroll
-  for (int nf = 0; nf < 2; ++nf) {
-    int n = n0 + wn0 + nf * 16 + fr;
-    if (n >= p.Cout) continue;
-    float bv = p.bias ? b2f(p.bias[n]) : 0.f;
-    #pragma unroll
-    for (int mf = 0; mf < 4; ++mf) {
-      #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int rl = wm0 + mf * 16 + fg * 4 + r;
-        if (!sm.rowok[rl]) continue;
-        float v = apply_act(acc[mf][nf][r] + bv, p.act, p.slope);
         p.y[sm.rowyb[rl] + n] = f2b(v);
       }
     }
@@ -1582,19 +1406,22 @@ static int conv_bn() {
   return v;
 }
 
-template <bool IS_CONVT, int STRIDE, int NBUF, int NW, int BNT>
+template <bool IS_CONVT, int STRIDE, int NBUF, int NW, int BNT,
+          bool PHASED = false>
 static void launch_one_glds(const ConvParams& p, dim3 grid,
                             hipStream_t stream) {
   constexpr size_t SMB = sizeof(ConvSmemT<NBUF, BNT>);
   static bool init = []() {
     hipFuncSetAttribute(
-        (const void*)(conv_glds_kernel<IS_CONVT, STRIDE, NBUF, NW, BNT>),
+        (const void*)(conv_glds_kernel<IS_CONVT, STRIDE, NBUF, NW, BNT,
+                                       PHASED>),
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)SMB);
     return true;
   }();
   (void)init;
-  hipLaunchKernelGGL((conv_glds_kernel<IS_CONVT, STRIDE, NBUF, NW, BNT>),
-                     grid, dim3(NW * 64), SMB, stream, p);
+  hipLaunchKernelGGL(
+      (conv_glds_kernel<IS_CONVT, STRIDE, NBUF, NW, BNT, PHASED>), grid,
+      dim3(NW * 64), SMB, stream, p);
 }
 
 template <bool IS_CONVT>
@@ -1644,8 +1471,17 @@ static void launch_conv(const ConvParams& p, bool is_convt, hipStream_t stream) 
     ConvParams q = p;
     long Mp = (long)q.B * (q.OH / 2) * (q.OW / 2);
     q.mtiles = 4 * cdiv(Mp, BM);
-    dim3 g2((long)q.mtiles * q.ntiles);
-    hipLaunchKernelGGL(convt_phased_kernel, g2, dim3(NTHREADS), 0, stream, q);
+    if (conv_nw() == 8 && conv_bn() == 128 && (q.Cout % 128) == 0) {
+      q.ntiles = (q.Cout + 127) / 128;
+      dim3 g2((long)q.mtiles * q.ntiles);
+      launch_one_glds<true, 2, 2, 8, 128, true>(q, g2, stream);
+    } else if (conv_nw() == 8) {
+      dim3 g2((long)q.mtiles * q.ntiles);
+      launch_one_glds<true, 2, 2, 8, 64, true>(q, g2, stream);
+    } else {
+      dim3 g2((long)q.mtiles * q.ntiles);
+      launch_one_glds<true, 2, 2, 4, 64, true>(q, g2, stream);
+    }
     return;
   }
   if (glds_ok) {

@@ -283,24 +283,6 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
   }
 }
 
-// grid (cdiv(C,256), nchunks): each block sums 8 sb-rows, atomics once.
-__global__ void in_bwd_dgb_kernel(const float* __restrict__ p1,
-                                  const float* __restrict__ p2,
-                                  float* __restrict__ dbeta,
-                                  float* __restrict__ dgamma, int SB, int C) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  int sb0 = blockIdx.y * 8;
-  int sb1 = min(sb0 + 8, SB);
-  float db = 0, dg = 0;
-  for (int sb = sb0; sb < sb1; ++sb) {
-    db += p1[(long)sb * C + c];
-    dg += p2[(long)sb * C + c];
-  }
-  atomicAdd(&dbeta[c], db);
-  atomicAdd(&dgamma[c], dg);
-}
-
 // ---- channel sum: out[c] = sum over (b,h,w) of x[...,c] (bias grads) ----
 __global__ __launch_bounds__(NT) void channel_sum_kernel(
     const short* __restrict__ x, float* __restrict__ out, long rows, int C,

@@ -1110,7 +1110,7 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
 
   // ---- per-lane gather identities ----
   // A: 16 instrs: instr i covers rows i*4..i*4+3; per-wave share = API
-  int a_row[API], a_k[API], a_tap[API], a_ci[API], a_dkh[API], a_dkw[API];
+  int a_row[API], a_ci[API], a_dkh[API], a_dkw[API];
   bool a_kv[API];
   #pragma unroll
   for (int j = 0; j < API; ++j) {
@@ -1123,7 +1123,6 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     a_ci[j] = (int)(k - (long)tap * p.Cin);
     a_dkh[j] = tap / p.KW;
     a_dkw[j] = tap - a_dkh[j] * p.KW;
-    a_k[j] = (int)(k - k0);
   }
   // D: instr i covers DRPI rows; per-wave share = DPI
   int d_row[DPI], d_n[DPI];

@@ -252,8 +252,10 @@ template <bool IS_CONVT, int STRIDE, int NBUF = 2, int NW = 4, int BNT = BN,
 __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
   constexpr int NT = NW * 64;        // threads
   constexpr int API = 16 / NW;       // A glds instructions per wave
-  constexpr int BPI = (BNT / 8) / NW;  // B glds instructions per wave
-  constexpr int NWC = BNT / 32;      // wave-grid columns (32-n wave tiles)
+  constexpr int TBI = BNT / 8;       // B glds instructions total
+  constexpr int BPI = TBI >= NW ? TBI / NW : 1;  // per wave (maybe idle)
+  constexpr int NF = BNT >= 32 ? 2 : 1;  // n-fragments per wave
+  constexpr int NWC = BNT / (NF * 16);   // wave-grid columns
   constexpr int NWR = NW / NWC;      // wave-grid rows
   constexpr int MF = (BM / NWR) / 16;  // m-fragments per wave
   const int stride = STRIDE ? STRIDE : p.stride;
@@ -336,7 +338,8 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
   int bn[BPI];
   #pragma unroll
   for (int j = 0; j < BPI; ++j) bn[j] = (w * BPI + j) * 8 + lr + n0;
-  static_assert(BNT == 64 || BNT == 128, "BNT");
+  const bool bw_on = w * BPI < TBI;  // waves beyond TBI stage no B
+  static_assert(BNT == 16 || BNT == 64 || BNT == 128, "BNT");
 
   auto rx = __builtin_amdgcn_make_buffer_rsrc(
       (void*)p.x, 0, (unsigned)((long)p.B * p.H * p.W * p.Cin * 2), 0x00020000);
@@ -454,18 +457,20 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
           rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * API + j) * 512],
           16, vo, 0, 0, 0);
     }
-    #pragma unroll
-    for (int j = 0; j < BPI; ++j) {
-      unsigned vo = (kv && bnv[j]) ? bvo[j] : 0xFF000000u;
-      __builtin_amdgcn_raw_ptr_buffer_load_lds(
-          rw, (__attribute__((address_space(3))) void*)&sm.Bt[buf][(w * BPI + j) * 512],
-          16, vo, 0, 0, 0);
+    if (bw_on) {
+      #pragma unroll
+      for (int j = 0; j < BPI; ++j) {
+        unsigned vo = (kv && bnv[j]) ? bvo[j] : 0xFF000000u;
+        __builtin_amdgcn_raw_ptr_buffer_load_lds(
+            rw, (__attribute__((address_space(3))) void*)&sm.Bt[buf][(w * BPI + j) * 512],
+            16, vo, 0, 0, 0);
+      }
     }
   };
 
-  v4f acc[MF][2] = {};
+  v4f acc[MF][NF] = {};
   const int wr = w / NWC, wc = w % NWC;
-  const int wm0 = wr * (MF * 16), wn0 = wc * 32;
+  const int wm0 = wr * (MF * 16), wn0 = wc * (NF * 16);
   const int fr = lane & 15;
   const int fg = lane >> 4;
   const int swz = (fr & 7) << 4;  // read-side XOR (bytes)
@@ -473,6 +478,17 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
   init_state();
   stage(0);
   __syncthreads();
+#ifdef CYG_DEBUG_EPI
+  if (bid == 0 && tid == 0)
+    printf("dbg NW=%d BNT=%d API=%d TBI=%d BPI=%d NF=%d NWC=%d MF=%d nk=%d "
+           "kv=%d klog=%d bvo0=%u bnv0=%d avo0=%u avalid0=%d "
+           "A0123=%d %d %d %d Bt0123=%d %d %d %d\n",
+           NW, BNT, API, TBI, BPI, NF, NWC, MF, nk, (int)kv, klog, bvo[0],
+           (int)bnv[0], avo[0], (int)avalid[0],
+           (int)sm.A[0][0], (int)sm.A[0][1], (int)sm.A[0][2], (int)sm.A[0][3],
+           (int)sm.Bt[0][0], (int)sm.Bt[0][1], (int)sm.Bt[0][2],
+           (int)sm.Bt[0][3]);
+#endif
 
   for (int kt = 0; kt < nk; ++kt) {
     if (kt + 1 < nk) {
@@ -484,26 +500,25 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
     #pragma unroll
     for (int kk = 0; kk < BK; kk += 32) {
       const int kbyte = (kk + fg * 8) * 2;
-      v8bf a0 = *(const v8bf*)(Ab + ((wm0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
-      v8bf a1 = *(const v8bf*)(Ab + ((wm0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
-      v8bf a2 = *(const v8bf*)(Ab + ((wm0 + 2 * 16 + fr) << 7) + (kbyte ^ swz));
-      v8bf a3 = *(const v8bf*)(Ab + ((wm0 + 3 * 16 + fr) << 7) + (kbyte ^ swz));
-      v8bf b0 = *(const v8bf*)(Bb + ((wn0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
-      v8bf b1 = *(const v8bf*)(Bb + ((wn0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
-      acc[2][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b0, acc[2][0], 0, 0, 0);
-      acc[2][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b1, acc[2][1], 0, 0, 0);
-      acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b0, acc[3][0], 0, 0, 0);
-      acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b1, acc[3][1], 0, 0, 0);
+      v8bf a[MF], b[NF];
+      #pragma unroll
+      for (int mf = 0; mf < MF; ++mf)
+        a[mf] = *(const v8bf*)(Ab + ((wm0 + mf * 16 + fr) << 7) + (kbyte ^ swz));
+      #pragma unroll
+      for (int nf = 0; nf < NF; ++nf)
+        b[nf] = *(const v8bf*)(Bb + ((wn0 + nf * 16 + fr) << 7) + (kbyte ^ swz));
+      #pragma unroll
+      for (int mf = 0; mf < MF; ++mf)
+        #pragma unroll
+        for (int nf = 0; nf < NF; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mf], b[nf], acc[mf][nf], 0, 0, 0);
     }
     __syncthreads();  // drains the in-flight glds (vmcnt(0) in the release)
   }
 
   #pragma unroll
-  for (int nf = 0; nf < 2; ++nf) {
+  for (int nf = 0; nf < NF; ++nf) {
     int n = n0 + wn0 + nf * 16 + fr;
     if (n >= p.Cout) continue;
     float bv = p.bias ? b2f(p.bias[n]) : 0.f;
@@ -1422,10 +1437,25 @@ static void launch_one_glds(const ConvParams& p, dim3 grid,
   hipLaunchKernelGGL(
       (conv_glds_kernel<IS_CONVT, STRIDE, NBUF, NW, BNT, PHASED>), grid,
       dim3(NW * 64), SMB, stream, p);
+  hipError_t err = hipGetLastError();
+  TORCH_CHECK(err == hipSuccess, "conv_glds launch failed (NW=", NW,
+              " BNT=", BNT, " PHASED=", PHASED, "): ",
+              hipGetErrorString(err));
 }
 
 template <bool IS_CONVT>
 static void launch_glds_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
+  if (p.Cout <= 16) {
+    // tiny-N (the 1-8 channel heads, padded to 8): 16-wide n-tiles
+    ConvParams q = p;
+    q.ntiles = (p.Cout + 15) / 16;
+    dim3 g2((long)q.mtiles * q.ntiles);
+    switch (p.stride) {
+      case 1: launch_one_glds<IS_CONVT, 1, 2, 4, 16>(q, g2, stream); return;
+      case 2: launch_one_glds<IS_CONVT, 2, 2, 4, 16>(q, g2, stream); return;
+      default: launch_one_glds<IS_CONVT, 0, 2, 4, 16>(q, g2, stream); return;
+    }
+  }
   if (conv_nw() == 8 && conv_bn() == 128 && (p.Cout % 128) == 0) {
     // wider n-tile: 1.5x arithmetic intensity for Cout >= 128 layers
     ConvParams q = p;

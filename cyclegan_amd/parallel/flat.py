@@ -43,6 +43,16 @@ class FlatParamGroup:
         for p, (off, k) in zip(self.params, self._offsets):
             p.grad = self.flat_grad[off:off + k].view(p.shape)
 
+    def set_grads(self, grads):
+        """Write a ``torch.autograd.grad`` result into the flat buffer.
+
+        One multi-tensor-apply copy per group instead of per-param
+        AccumulateGrad adds; every view is fully overwritten, so no
+        zero-fill between steps is needed."""
+        torch._foreach_copy_(
+            [p.grad for p in self.params],
+            [g.reshape(p.shape) for p, g in zip(self.params, grads)])
+
     def zero_grad(self):
         self.flat_grad.zero_()
         # autograd accumulates in-place into existing .grad views; re-attach

@@ -94,9 +94,6 @@ class CycleGAN:
 
     def train_step(self, x, y) -> Dict[str, torch.Tensor]:
         x, y = self._cast(x), self._cast(y)
-        for g in self.groups.values():
-            g.zero_grad()
-
         b = x.shape[0]
         # batched generator calls: every op is per-sample (convs, per-sample
         # InstanceNorm stats, per-sample losses), so G(cat(x,y)) is
@@ -125,14 +122,22 @@ class CycleGAN:
         X_loss = self.discriminator_loss(dx[:b], dx[b:])
         Y_loss = self.discriminator_loss(dy_[:b], dy_[b:])
 
-        torch.autograd.backward(G_total, inputs=self.groups["G"].params,
-                                retain_graph=True)
+        # torch.autograd.grad (not .backward): grads come back as fresh
+        # tensors and land in the flat buffer via ONE batched multi-tensor
+        # copy per group, skipping AccumulateGrad's per-param add into the
+        # .grad views (~300 tiny launches/step) and the flat zero-fill.
+        self.groups["G"].set_grads(
+            torch.autograd.grad(G_total, self.groups["G"].params,
+                                retain_graph=True))
         self.sync.launch(self.groups["G"].flat_grad)
-        torch.autograd.backward(F_total, inputs=self.groups["F"].params)
+        self.groups["F"].set_grads(
+            torch.autograd.grad(F_total, self.groups["F"].params))
         self.sync.launch(self.groups["F"].flat_grad)
-        torch.autograd.backward(X_loss, inputs=self.groups["X"].params)
+        self.groups["X"].set_grads(
+            torch.autograd.grad(X_loss, self.groups["X"].params))
         self.sync.launch(self.groups["X"].flat_grad)
-        torch.autograd.backward(Y_loss, inputs=self.groups["Y"].params)
+        self.groups["Y"].set_grads(
+            torch.autograd.grad(Y_loss, self.groups["Y"].params))
         self.sync.launch(self.groups["Y"].flat_grad)
         self.sync.wait_all()
 

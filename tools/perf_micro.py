@@ -74,6 +74,18 @@ cases += [
     ("stem wgrad  ", fl4, lambda: e.conv2d_wgrad(x4, mk(B, 256, 256, 64), 7, 7, 1, 3, 3, True)),
 ]
 
+# generator head 7x7 64->3 (padded to 8) @256^2, reflect pad 3 (BN16 tile path)
+x5 = mk(B, 256, 256, 64)
+w5 = mk(8, 7, 7, 64) * 0.1
+wt5 = w5.permute(3, 1, 2, 0).contiguous()
+dy5 = mk(B, 256, 256, 8)
+fl5 = 2 * (B * 256 * 256) * 8 * (49 * 64)
+cases += [
+    ("head fwd    ", fl5, lambda: e.conv2d_fwd(x5, w5, None, 1, 3, 3, 3, 3, True, 0, 0.2)),
+    ("head dgrad  ", fl5, lambda: e.conv2d_dgrad(dy5, wt5, 256, 256, 1, 3, 3, 3, 3, True)),
+    ("head wgrad  ", fl5, lambda: e.conv2d_wgrad(x5, dy5, 7, 7, 1, 3, 3, True)),
+]
+
 # InstanceNorm 64^2 x 256
 g = torch.rand(256, device=DEV)
 bta = torch.rand(256, device=DEV)
@@ -87,7 +99,7 @@ cases += [
 mean_ = torch.rand(B, 256, device=DEV)
 rstd_ = torch.rand(B, 256, device=DEV) + 0.5
 cases += [
-    ("IN bwd      ", None, lambda: e.instnorm_bwd(dy, xin, g, mean_, rstd_)),
+    ("IN bwd      ", None, lambda: e.instnorm_bwd(dy, xin, g, mean_, rstd_, None, 0, 0.2)),
 ]
 
 for name, fl, fn in cases:

@@ -47,6 +47,11 @@ class FusedAdam:
         if backend.use_hip(self.p):
             backend.ext().adam_step(self.p, self.g, self.m, self.v,
                                     self.lr, self.b1, self.b2, self.eps, self.t)
+            # the raw kernel bypasses the dispatcher, so the in-place update
+            # does not bump the version counter that invalidates the bf16
+            # shadow-weight caches (ops.shadow) — bump it explicitly, or
+            # compute would keep running on the step-0 weights forever
+            torch.autograd.graph.increment_version(self.p)
             return
         lr_t = self.lr * math.sqrt(1 - self.b2 ** self.t) / (1 - self.b1 ** self.t)
         self.m.mul_(self.b1).add_(self.g, alpha=1 - self.b1)

@@ -43,6 +43,16 @@ class FlatParamGroup:
         for p, (off, k) in zip(self.params, self._offsets):
             p.grad = self.flat_grad[off:off + k].view(p.shape)
 
+    def bump_versions(self):
+        """Invalidate per-param version counters after an in-place flat
+        update that bypassed the dispatcher (the fused Adam kernel).
+
+        ``p.data = view`` keeps each param's OWN version counter — bumping
+        the flat buffer does not propagate — so the bf16 shadow caches
+        (ops.shadow, keyed on param version) would never refresh."""
+        for p in self.params:
+            torch.autograd.graph.increment_version(p)
+
     def set_grads(self, grads):
         """Write a ``torch.autograd.grad`` result into the flat buffer.
 

@@ -143,6 +143,8 @@ class CycleGAN:
 
         for opt in self.optimizers.values():
             opt.step()
+        for g in self.groups.values():
+            g.bump_versions()  # invalidate bf16 shadow caches (see flat.py)
 
         return {
             "loss_G/loss": G_loss.detach(), "loss_G/cycle": G_cycle_loss.detach(),

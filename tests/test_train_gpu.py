@@ -39,6 +39,27 @@ def test_train_step_bf16(tmp_path):
         assert g.check_views()
 
 
+def test_shadow_weights_refresh_after_step(tmp_path):
+    """The fused Adam bypasses the dispatcher; it must still invalidate the
+    bf16 shadow caches so compute sees the updated masters (regression:
+    shadows froze at step 0 while masters trained)."""
+    from cyclegan_amd.parallel import DistContext
+    from cyclegan_amd.trainer import CycleGAN
+    from cyclegan_amd.ops import shadow
+    torch.manual_seed(0)
+    ctx = DistContext(device=torch.device("cuda", 0))
+    gan = CycleGAN(make_args(tmp_path), ctx)
+    w = next(p for p in gan.G.parameters() if p.dim() == 4)
+    like = torch.empty(1, device=ctx.device, dtype=torch.bfloat16)
+    before = shadow.compute_weight(w, like).clone()
+    x = torch.rand(2, 64, 64, 3, device=ctx.device, dtype=torch.bfloat16)
+    gan.train_step(x, x)
+    torch.cuda.synchronize()
+    after = shadow.compute_weight(w, like)
+    assert not torch.equal(before, after), "bf16 shadow did not refresh"
+    assert torch.equal(after, w.detach().to(torch.bfloat16))
+
+
 def test_gpu_step_matches_cpu_fp32_closely(tmp_path):
     """One step, same weights+data: bf16 HIP losses must track the fp32 CPU
     reference losses to bf16 tolerance (catches systematic kernel bias)."""

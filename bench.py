@@ -76,11 +76,19 @@ def main():
 
     for i in range(args.warmup):
         gan.train_step(*pool[i % len(pool)])
+
+    # steady-state step as ONE hip graph (N=1; RCCL capture not enabled
+    # yet) — every replay still copies the step's input batch in and runs
+    # the full forward/backward/optimizer.
+    step = gan.train_step
+    if on_gpu and ctx.world_size == 1 and not os.environ.get("CYG_NO_GRAPH"):
+        from cyclegan_amd.trainer import GraphedStep
+        step = GraphedStep(gan, *pool[0])
     sync()
 
     t0 = time.perf_counter()
     for i in range(args.steps):
-        gan.train_step(*pool[i % len(pool)])
+        step(*pool[i % len(pool)])
     if on_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0

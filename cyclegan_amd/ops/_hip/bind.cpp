@@ -36,6 +36,8 @@ at::Tensor persample_loss_const_fwd(at::Tensor, double, bool);
 std::vector<at::Tensor> persample_loss_bwd(at::Tensor, at::Tensor, at::Tensor,
                                            bool, bool, bool);
 at::Tensor persample_loss_const_bwd(at::Tensor, double, at::Tensor, bool);
+void adam_step_dev(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                   double, double, double);
 void adam_step(at::Tensor, at::Tensor, at::Tensor, at::Tensor, double, double,
                double, double, int64_t);
 }  // namespace cyg
@@ -62,4 +64,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("persample_loss_bwd", &cyg::persample_loss_bwd);
   m.def("persample_loss_const_bwd", &cyg::persample_loss_const_bwd);
   m.def("adam_step", &cyg::adam_step);
+  m.def("adam_step_dev", &cyg::adam_step_dev);
 }

@@ -481,6 +481,26 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
   }
 }
 
+// graph-capture variant: lr_t read from device memory so a captured step
+// picks up the per-step bias correction the host writes before each replay
+__global__ void adam_kernel_dev(float* __restrict__ p,
+                                const float* __restrict__ g,
+                                float* __restrict__ m, float* __restrict__ v,
+                                long n, const float* __restrict__ lr_t_ptr,
+                                float b1, float b2, float eps) {
+  const float lr_t = *lr_t_ptr;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float gi = g[i];
+    float mi = b1 * m[i] + (1.f - b1) * gi;
+    float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    p[i] -= lr_t * mi / (sqrtf(vi) + eps);
+  }
+}
+
 // ================= host wrappers =================
 
 static int slices_for(long HW, int B) {
@@ -704,6 +724,22 @@ void adam_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                      (const float*)g.const_data_ptr(),
                      (float*)m.mutable_data_ptr(),
                      (float*)v.mutable_data_ptr(), n, (float)lr_t, (float)b1,
+                     (float)b2, (float)eps);
+}
+
+void adam_step_dev(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
+                   at::Tensor lr_t, double b1, double b2, double eps) {
+  TORCH_CHECK(p.is_cuda() && p.scalar_type() == at::kFloat);
+  TORCH_CHECK(lr_t.is_cuda() && lr_t.scalar_type() == at::kFloat);
+  long n = p.numel();
+  int blocks = (int)std::min<long>(2048, (n + 255) / 256);
+  hipLaunchKernelGGL(adam_kernel_dev, dim3(blocks), dim3(256), 0,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     (float*)p.mutable_data_ptr(),
+                     (const float*)g.const_data_ptr(),
+                     (float*)m.mutable_data_ptr(),
+                     (float*)v.mutable_data_ptr(), n,
+                     (const float*)lr_t.const_data_ptr(), (float)b1,
                      (float)b2, (float)eps);
 }
 

@@ -40,9 +40,43 @@ class FusedAdam:
         self.m = torch.zeros_like(flat_param)
         self.v = torch.zeros_like(flat_param)
         self.t = 0
+        # graph-capture mode: the update kernel reads lr_t from this device
+        # scalar instead of an immediate, so a captured step follows the
+        # per-step bias correction written by advance_lr() before replay
+        self.graph_mode = False
+        self._lr_t_dev = None
+
+    def _lr_t(self) -> float:
+        return (self.lr * math.sqrt(1 - self.b2 ** self.t)
+                / (1 - self.b1 ** self.t))
+
+    @torch.no_grad()
+    def prepare_graph(self):
+        """Switch to the capture-safe kernel and allocate the device lr_t
+        scalar (value is a placeholder: capture records, never executes)."""
+        self.graph_mode = True
+        if self._lr_t_dev is None:
+            self._lr_t_dev = torch.zeros(1, dtype=torch.float32,
+                                         device=self.p.device)
+
+    @torch.no_grad()
+    def advance_lr(self):
+        """t += 1 and refresh the device lr_t scalar (graph mode: called
+        once per step OUTSIDE the captured region)."""
+        self.t += 1
+        if self._lr_t_dev is None:
+            self._lr_t_dev = torch.empty(1, dtype=torch.float32,
+                                         device=self.p.device)
+        self._lr_t_dev.fill_(self._lr_t())
 
     @torch.no_grad()
     def step(self):
+        if self.graph_mode:
+            # t/lr_t advanced by advance_lr(); capture-safe kernel
+            backend.ext().adam_step_dev(self.p, self.g, self.m, self.v,
+                                        self._lr_t_dev, self.b1, self.b2,
+                                        self.eps)
+            return
         self.t += 1
         if backend.use_hip(self.p):
             backend.ext().adam_step(self.p, self.g, self.m, self.v,

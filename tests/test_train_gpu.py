@@ -98,3 +98,44 @@ def test_checkpoint_roundtrip_gpu(tmp_path):
     gan2 = CycleGAN(make_args(tmp_path), ctx)
     assert gan2.load_checkpoint()
     assert torch.equal(gan.groups["G"].flat_param, gan2.groups["G"].flat_param)
+
+
+def test_graphed_step_matches_eager(tmp_path):
+    """4 logical steps, eager vs (2 eager warmup + 2 graph replays): the
+    captured step must track the eager trajectory (catches frozen lr_t,
+    missing shadow recasts, or stale inputs inside the graph)."""
+    from cyclegan_amd.parallel import DistContext
+    from cyclegan_amd.trainer import CycleGAN, GraphedStep
+    ctx = DistContext(device=torch.device("cuda", 0))
+    data = []
+    g = torch.Generator().manual_seed(11)
+    for _ in range(4):
+        data.append((torch.rand(2, 64, 64, 3, generator=g) * 2 - 1,
+                     torch.rand(2, 64, 64, 3, generator=g) * 2 - 1))
+
+    torch.manual_seed(3)
+    gan_e = CycleGAN(make_args(tmp_path), ctx)
+    for x, y in data:
+        r_e = gan_e.train_step(x, y)
+    torch.cuda.synchronize()
+
+    torch.manual_seed(3)
+    gan_g = CycleGAN(make_args(tmp_path), ctx)
+    for x, y in data[:2]:
+        gan_g.train_step(x, y)
+    step = GraphedStep(gan_g, *data[2], warmup=0)
+    for x, y in data[2:]:
+        r_g = step(x, y)
+    torch.cuda.synchronize()
+
+    assert all(o.t == 4 for o in gan_g.optimizers.values())
+    for k in r_e:
+        a, b = r_e[k].item(), r_g[k].item()
+        assert abs(a - b) <= 0.03 * (abs(b) + 0.03), (k, a, b)
+    # atomic-order nondeterminism (IN dgamma/dbeta) perturbs grads in the
+    # last ulp; Adam's sqrt(v) normalization can then move single params by
+    # ~lr_t per step, so two EAGER runs differ by up to a few lr_t too —
+    # the bound only needs to catch systematic divergence (stale shadows,
+    # frozen lr_t), which shows up at 1e-2+.
+    d = (gan_e.groups["G"].flat_param - gan_g.groups["G"].flat_param)
+    assert d.abs().max().item() < 2.5e-3

@@ -43,6 +43,15 @@ struct ConvParams {
   int mtiles, ntiles;
 };
 
+// XCD-aware block remap: hardware dispatches blockIdx round-robin over the
+// 8 XCDs, so consecutive tiles (which gather overlapping input rows) land
+// in different XCDs' L2s and each L2 re-pulls the same lines from HBM.
+// Chunking gives every XCD a CONTIGUOUS tile range: XCD k executes blocks
+// k, k+8, k+16, ... which remap to tiles k*nb/8 + 0, 1, 2, ...
+DEV int xcd_chunk(int bid, int nb) {
+  return (nb % 8 == 0) ? (bid % 8) * (nb / 8) + bid / 8 : bid;
+}
+
 // ---------------- shared GEMM core ----------------
 // As rows = m (output pixel), Bs rows = n (cout); both k-contiguous.
 
@@ -59,7 +68,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_gemm_kernel(ConvParams p) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int bid = blockIdx.x;
+  const int bid = xcd_chunk(blockIdx.x, gridDim.x);
   const int mt = bid % p.mtiles, nt = bid / p.mtiles;
   const long m0 = (long)mt * BM;
   const int n0 = nt * BN;
@@ -266,7 +275,11 @@ __global__ __launch_bounds__(NW * 64) void conv_glds_kernel(ConvParams p) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int w = tid >> 6;
-  const int bid = blockIdx.x;
+  // stride-2 convT-gather decode prefers the hardware round-robin (its
+  // taps scatter anyway and chunking hotspots DRAM channels): measured
+  // -25% on the s2 dgrad shape, +6% on the K3 forward.
+  const int bid = (IS_CONVT && STRIDE == 2)
+                      ? blockIdx.x : xcd_chunk(blockIdx.x, gridDim.x);
   const int mt = bid % p.mtiles, nt = bid / p.mtiles;
   const int n0 = nt * BNT;
 
@@ -593,7 +606,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_fp8_kernel(ConvParams p) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int w = tid >> 6;
-  const int bid = blockIdx.x;
+  const int bid = xcd_chunk(blockIdx.x, gridDim.x);
   const int mt = bid % p.mtiles, nt = bid / p.mtiles;
   const long m0 = (long)mt * BM;
   const int n0 = nt * BN;
@@ -810,9 +823,10 @@ __global__ __launch_bounds__(NTHREADS) void wgrad_kernel(WgradParams p) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int kt = blockIdx.x % p.ktiles;
-  const int nt = (blockIdx.x / p.ktiles) % p.ntiles;
-  const int sl = blockIdx.x / (p.ktiles * p.ntiles);
+  const int wbid = xcd_chunk(blockIdx.x, gridDim.x);
+  const int kt = wbid % p.ktiles;
+  const int nt = (wbid / p.ktiles) % p.ntiles;
+  const int sl = wbid / (p.ktiles * p.ntiles);
   const long k0 = (long)kt * WG_BK;
   const int n0 = nt * WG_BN;
   const long mstart = sl * p.mchunks_per_slice * WG_BM;
@@ -1107,9 +1121,10 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int w = tid >> 6;
-  const int kt = blockIdx.x % p.ktiles;
-  const int nt = (blockIdx.x / p.ktiles) % p.ntiles;
-  const int sl = blockIdx.x / (p.ktiles * p.ntiles);
+  const int wbid = xcd_chunk(blockIdx.x, gridDim.x);
+  const int kt = wbid % p.ktiles;
+  const int nt = (wbid / p.ktiles) % p.ntiles;
+  const int sl = wbid / (p.ktiles * p.ntiles);
   const long k0 = (long)kt * WG_BK;
   const int n0 = nt * WBN;
   const long mstart = sl * p.mchunks_per_slice * WG_BM;

@@ -1339,9 +1339,18 @@ __global__ void wgrad_reduce_kernel(const float* __restrict__ ws,
   if (k >= KTOT || n >= Cout) return;
   long stride = (long)ktiles * ntiles * wbn * WG_BK;
   long off = (((long)kt * ntiles + nt) * wbn + nl) * WG_BK + kl;
-  float s = 0;
-  for (int sl = 0; sl < slices; ++sl) s += ws[off + sl * stride];
-  dw[(long)n * KTOT + k] = s;
+  // 4 independent accumulation chains: with ~14 slices per element the
+  // single-chain loop was latency-bound at 4x the memory floor
+  float s0 = 0, s1 = 0, s2 = 0, s3 = 0;
+  int sl = 0;
+  for (; sl + 4 <= slices; sl += 4) {
+    s0 += ws[off + (sl + 0) * stride];
+    s1 += ws[off + (sl + 1) * stride];
+    s2 += ws[off + (sl + 2) * stride];
+    s3 += ws[off + (sl + 3) * stride];
+  }
+  for (; sl < slices; ++sl) s0 += ws[off + sl * stride];
+  dw[(long)n * KTOT + k] = (s0 + s1) + (s2 + s3);
 }
 
 // ---------------- reflect fold (dgrad border scatter) ----------------

@@ -15,4 +15,4 @@ from . import backend  # noqa: F401
 from .conv import conv2d, conv_transpose2d, same_pads, set_fp8_mode, fp8_mode  # noqa: F401
 from .norm import instance_norm  # noqa: F401
 from .pad import reflection_pad2d  # noqa: F401
-from .losses import MAE, MSE, MSE_const  # noqa: F401
+from .losses import MAE, MSE, MSE_const, BCE  # noqa: F401

@@ -82,3 +82,20 @@ def MSE_const(y_pred: torch.Tensor, const: float) -> torch.Tensor:
         return _PerSampleConstFn.apply(y_pred, const, True)
     d = y_pred.float() - const
     return (d * d).mean(dim=tuple(range(1, d.dim())))
+
+
+def BCE(y_true: torch.Tensor, y_pred: torch.Tensor) -> torch.Tensor:
+    """Per-sample binary cross-entropy on probabilities, shape [B] fp32.
+
+    Parity with the reference's (dead — never called) BCE helper
+    (/root/reference/main.py:98-103, tf.keras.losses.binary_crossentropy
+    with from_logits=False): mean over non-batch axes of
+    -[t·log(p) + (1-t)·log(1-p)], with TF's probability clamp to
+    [eps, 1-eps], eps=1e-7. Kept for component-inventory completeness;
+    the LSGAN objective uses MSE instead.
+    """
+    eps = 1e-7
+    p = y_pred.float().clamp(eps, 1.0 - eps)
+    t = y_true.float()
+    v = -(t * p.log() + (1.0 - t) * torch.log1p(-p))
+    return v.mean(dim=tuple(range(1, v.dim())))

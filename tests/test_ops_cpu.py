@@ -177,3 +177,20 @@ def test_gradcheck_instance_norm_small():
     assert torch.autograd.gradcheck(
         lambda a, gg, bb: ops.instance_norm(a, gg, bb, eps=1e-3), (x, g, b),
         eps=1e-6, atol=1e-4)
+
+
+def test_bce_matches_tf_semantics():
+    """BCE (reference dead-code parity): per-sample mean, prob clamp."""
+    import torch
+    from cyclegan_amd.ops import BCE
+    g = torch.Generator().manual_seed(0)
+    p = torch.rand(3, 5, 5, 2, generator=g)
+    t = (torch.rand(3, 5, 5, 2, generator=g) > 0.5).float()
+    out = BCE(t, p)
+    assert out.shape == (3,)
+    ref = torch.nn.functional.binary_cross_entropy(
+        p, t, reduction="none").mean(dim=(1, 2, 3))
+    assert torch.allclose(out, ref, atol=1e-5)
+    # clamp keeps exact 0/1 predictions finite (TF semantics)
+    assert torch.isfinite(BCE(torch.ones(1, 4), torch.ones(1, 4))).all()
+    assert torch.isfinite(BCE(torch.ones(1, 4), torch.zeros(1, 4))).all()

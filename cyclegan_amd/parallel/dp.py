@@ -80,18 +80,36 @@ class DistContext:
 
 class GradSync:
     """Issues one async flat all-reduce per model group, waited before the
-    optimizer steps."""
+    optimizer steps.
 
-    def __init__(self, ctx: DistContext):
+    With ``timing=True`` (CUDA only) each all-reduce is bracketed by hip
+    events on the comm stream; ``pop_comm_ms()`` returns the accumulated
+    RCCL time since the last read (one host sync per read — call it at
+    epoch granularity, not per step)."""
+
+    def __init__(self, ctx: DistContext, timing: bool = False):
         self.ctx = ctx
         self._pending = []
+        self._timer = None
+        if timing and ctx.device.type == "cuda":
+            from ..utils.profiler import HipEventTimer
+            self._timer = HipEventTimer()
 
     def launch(self, flat_grad: torch.Tensor):
+        if self._timer is not None and self.ctx.distributed:
+            self._timer.begin()
         h = self.ctx.all_reduce_async(flat_grad)
         if h is not None:
             self._pending.append(h)
+        if self._timer is not None and self.ctx.distributed:
+            self._timer.end()
 
     def wait_all(self):
         for h in self._pending:
             h.wait()
         self._pending.clear()
+
+    def pop_comm_ms(self) -> float:
+        """Accumulated all-reduce GPU time (ms) since last call; 0 if
+        timing is off or single-rank."""
+        return self._timer.elapsed_ms() if self._timer is not None else 0.0

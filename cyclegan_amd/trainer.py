@@ -66,7 +66,7 @@ class CycleGAN:
                        (("G", self.G), ("F", self.F), ("X", self.X), ("Y", self.Y))}
         self.optimizers = {name: FusedAdam(g.flat_param, g.flat_grad)
                            for name, g in self.groups.items()}
-        self.sync = GradSync(ctx)
+        self.sync = GradSync(ctx, timing=getattr(args, "verbose", 1) == 2)
 
     # ---- loss functions (reference main.py:172-195) ----
 

@@ -93,6 +93,11 @@ def main(args):
         end = time()
         if summary is not None:
             summary.scalar("elapse", end - start, step=epoch)
+            if args.verbose == 2:
+                # extension: RCCL all-reduce GPU-time per epoch (hip events
+                # around each flat-grad all-reduce, parallel/dp.py)
+                summary.scalar("comm/all_reduce_ms", gan.sync.pop_comm_ms(),
+                               step=epoch)
 
         if ctx.is_main and results:
             # (the reference console print swaps two labels, main.py:394-397

@@ -1421,14 +1421,6 @@ static void launch_conv_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
   }
 }
 
-static bool use_pipe3() {
-  static int v = []() {
-    const char* e = getenv("CYG_CONV_PIPE3");
-    return e ? atoi(e) : 0;
-  }();
-  return v != 0;
-}
-
 static int conv_nw() {
   static int v = []() {
     const char* e = getenv("CYG_CONV_NW");
@@ -1496,13 +1488,6 @@ static void launch_glds_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
       case 1: launch_one_glds<IS_CONVT, 1, 2, 8, 64>(p, grid, stream); return;
       case 2: launch_one_glds<IS_CONVT, 2, 2, 8, 64>(p, grid, stream); return;
       default: launch_one_glds<IS_CONVT, 0, 2, 8, 64>(p, grid, stream); return;
-    }
-  }
-  if (use_pipe3()) {
-    switch (p.stride) {
-      case 1: launch_one_glds<IS_CONVT, 1, 3, 8, 64>(p, grid, stream); return;
-      case 2: launch_one_glds<IS_CONVT, 2, 3, 8, 64>(p, grid, stream); return;
-      default: launch_one_glds<IS_CONVT, 0, 3, 8, 64>(p, grid, stream); return;
     }
   }
   switch (p.stride) {

@@ -88,34 +88,82 @@ __global__ __launch_bounds__(NT) void in_norm_kernel(
   const float inv_hw = 1.f / (float)HW;
 
   __shared__ float sm[MAXC], sr[MAXC];
-  for (int c = tid; c < C; c += NT) {
+  __shared__ float part[2 * NT];
+  if (C >= NT) {
+    for (int c = tid; c < C; c += NT) {
+      float sv = 0, qv = 0;
+      int k = 0;
+      for (; k + 4 <= S; k += 4) {  // 4 independent chains hide L2 latency
+        float s0 = psum[((long)(k + 0) * B + b) * C + c];
+        float s1 = psum[((long)(k + 1) * B + b) * C + c];
+        float s2 = psum[((long)(k + 2) * B + b) * C + c];
+        float s3 = psum[((long)(k + 3) * B + b) * C + c];
+        float q0 = psq[((long)(k + 0) * B + b) * C + c];
+        float q1 = psq[((long)(k + 1) * B + b) * C + c];
+        float q2 = psq[((long)(k + 2) * B + b) * C + c];
+        float q3 = psq[((long)(k + 3) * B + b) * C + c];
+        sv += (s0 + s1) + (s2 + s3);
+        qv += (q0 + q1) + (q2 + q3);
+      }
+      for (; k < S; ++k) {
+        sv += psum[((long)k * B + b) * C + c];
+        qv += psq[((long)k * B + b) * C + c];
+      }
+      float m = sv * inv_hw;
+      float var = qv * inv_hw - m * m;
+      if (var < 0.f) var = 0.f;
+      float rs = rsqrtf(var + eps);
+      sm[c] = m;
+      sr[c] = rs;
+      if (sl == 0) {
+        mean[(long)b * C + c] = m;
+        rstd[(long)b * C + c] = rs;
+      }
+    }
+  } else {
+    // narrow C (stem/upsample norms, C=64): split the slab sum over
+    // (c, slab-chunk) so all NT lanes work instead of C of them; the
+    // per-chunk partials meet in LDS (C and NT are powers of two)
+    const int PS = NT / C;
+    const int kc = tid / C;
+    const int c = tid - kc * C;
     float sv = 0, qv = 0;
-    int k = 0;
-    for (; k + 4 <= S; k += 4) {  // 4 independent chains hide L2 latency
-      float s0 = psum[((long)(k + 0) * B + b) * C + c];
-      float s1 = psum[((long)(k + 1) * B + b) * C + c];
-      float s2 = psum[((long)(k + 2) * B + b) * C + c];
-      float s3 = psum[((long)(k + 3) * B + b) * C + c];
-      float q0 = psq[((long)(k + 0) * B + b) * C + c];
-      float q1 = psq[((long)(k + 1) * B + b) * C + c];
-      float q2 = psq[((long)(k + 2) * B + b) * C + c];
-      float q3 = psq[((long)(k + 3) * B + b) * C + c];
+    int k = kc;
+    for (; k + 3 * PS < S; k += 4 * PS) {
+      float s0 = psum[((long)(k + 0 * PS) * B + b) * C + c];
+      float s1 = psum[((long)(k + 1 * PS) * B + b) * C + c];
+      float s2 = psum[((long)(k + 2 * PS) * B + b) * C + c];
+      float s3 = psum[((long)(k + 3 * PS) * B + b) * C + c];
+      float q0 = psq[((long)(k + 0 * PS) * B + b) * C + c];
+      float q1 = psq[((long)(k + 1 * PS) * B + b) * C + c];
+      float q2 = psq[((long)(k + 2 * PS) * B + b) * C + c];
+      float q3 = psq[((long)(k + 3 * PS) * B + b) * C + c];
       sv += (s0 + s1) + (s2 + s3);
       qv += (q0 + q1) + (q2 + q3);
     }
-    for (; k < S; ++k) {
+    for (; k < S; k += PS) {
       sv += psum[((long)k * B + b) * C + c];
       qv += psq[((long)k * B + b) * C + c];
     }
-    float m = sv * inv_hw;
-    float var = qv * inv_hw - m * m;
-    if (var < 0.f) var = 0.f;
-    float rs = rsqrtf(var + eps);
-    sm[c] = m;
-    sr[c] = rs;
-    if (sl == 0) {
-      mean[(long)b * C + c] = m;
-      rstd[(long)b * C + c] = rs;
+    part[tid] = sv;
+    part[NT + tid] = qv;
+    __syncthreads();
+    if (tid < C) {
+      float svt = 0, qvt = 0;
+      for (int j = 0; j < PS; ++j) {
+        svt += part[j * C + tid];
+        qvt += part[NT + j * C + tid];
+      }
+      float m = svt * inv_hw;
+      float var = qvt * inv_hw - m * m;
+      if (var < 0.f) var = 0.f;
+      float rs = rsqrtf(var + eps);
+      sm[tid] = m;
+      sr[tid] = rs;
+      if (sl == 0) {
+        mean[(long)b * C + tid] = m;
+        rstd[(long)b * C + tid] = rs;
+      }
     }
   }
   __syncthreads();
@@ -230,32 +278,76 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
   const float inv_hw = 1.f / (float)HW;
 
   __shared__ float sm1[MAXC], sm2[MAXC], smean[MAXC], srstd[MAXC];
-  for (int c = tid; c < C; c += NT) {
+  __shared__ float part[2 * NT];
+  if (C >= NT) {
+    for (int c = tid; c < C; c += NT) {
+      float t1 = 0, t2 = 0;
+      int k = 0;
+      for (; k + 4 <= S; k += 4) {
+        float a0 = p1[((long)(k + 0) * B + b) * C + c];
+        float a1 = p1[((long)(k + 1) * B + b) * C + c];
+        float a2 = p1[((long)(k + 2) * B + b) * C + c];
+        float a3 = p1[((long)(k + 3) * B + b) * C + c];
+        float b0 = p2[((long)(k + 0) * B + b) * C + c];
+        float b1 = p2[((long)(k + 1) * B + b) * C + c];
+        float b2 = p2[((long)(k + 2) * B + b) * C + c];
+        float b3 = p2[((long)(k + 3) * B + b) * C + c];
+        t1 += (a0 + a1) + (a2 + a3);
+        t2 += (b0 + b1) + (b2 + b3);
+      }
+      for (; k < S; ++k) {
+        t1 += p1[((long)k * B + b) * C + c];
+        t2 += p2[((long)k * B + b) * C + c];
+      }
+      sm1[c] = t1 * inv_hw;
+      sm2[c] = t2 * inv_hw;
+      smean[c] = mean[(long)b * C + c];
+      srstd[c] = rstd[(long)b * C + c];
+      if (sl == 0) {  // fold the dgamma/dbeta reduction in (B adders per c)
+        atomicAdd(&dbeta[c], t1);
+        atomicAdd(&dgamma[c], t2);
+      }
+    }
+  } else {
+    // narrow C: split the slab sum over (c, slab-chunk) — see in_norm
+    const int PS = NT / C;
+    const int kc = tid / C;
+    const int c = tid - kc * C;
     float t1 = 0, t2 = 0;
-    int k = 0;
-    for (; k + 4 <= S; k += 4) {
-      float a0 = p1[((long)(k + 0) * B + b) * C + c];
-      float a1 = p1[((long)(k + 1) * B + b) * C + c];
-      float a2 = p1[((long)(k + 2) * B + b) * C + c];
-      float a3 = p1[((long)(k + 3) * B + b) * C + c];
-      float b0 = p2[((long)(k + 0) * B + b) * C + c];
-      float b1 = p2[((long)(k + 1) * B + b) * C + c];
-      float b2 = p2[((long)(k + 2) * B + b) * C + c];
-      float b3 = p2[((long)(k + 3) * B + b) * C + c];
+    int k = kc;
+    for (; k + 3 * PS < S; k += 4 * PS) {
+      float a0 = p1[((long)(k + 0 * PS) * B + b) * C + c];
+      float a1 = p1[((long)(k + 1 * PS) * B + b) * C + c];
+      float a2 = p1[((long)(k + 2 * PS) * B + b) * C + c];
+      float a3 = p1[((long)(k + 3 * PS) * B + b) * C + c];
+      float b0 = p2[((long)(k + 0 * PS) * B + b) * C + c];
+      float b1 = p2[((long)(k + 1 * PS) * B + b) * C + c];
+      float b2 = p2[((long)(k + 2 * PS) * B + b) * C + c];
+      float b3 = p2[((long)(k + 3 * PS) * B + b) * C + c];
       t1 += (a0 + a1) + (a2 + a3);
       t2 += (b0 + b1) + (b2 + b3);
     }
-    for (; k < S; ++k) {
+    for (; k < S; k += PS) {
       t1 += p1[((long)k * B + b) * C + c];
       t2 += p2[((long)k * B + b) * C + c];
     }
-    sm1[c] = t1 * inv_hw;
-    sm2[c] = t2 * inv_hw;
-    smean[c] = mean[(long)b * C + c];
-    srstd[c] = rstd[(long)b * C + c];
-    if (sl == 0) {  // fold the dgamma/dbeta reduction in (B adders per c)
-      atomicAdd(&dbeta[c], t1);
-      atomicAdd(&dgamma[c], t2);
+    part[tid] = t1;
+    part[NT + tid] = t2;
+    __syncthreads();
+    if (tid < C) {
+      float s1 = 0, s2 = 0;
+      for (int j = 0; j < PS; ++j) {
+        s1 += part[j * C + tid];
+        s2 += part[NT + j * C + tid];
+      }
+      sm1[tid] = s1 * inv_hw;
+      sm2[tid] = s2 * inv_hw;
+      smean[tid] = mean[(long)b * C + tid];
+      srstd[tid] = rstd[(long)b * C + tid];
+      if (sl == 0) {
+        atomicAdd(&dbeta[tid], s1);
+        atomicAdd(&dgamma[tid], s2);
+      }
     }
   }
   __syncthreads();

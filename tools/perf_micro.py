@@ -71,7 +71,10 @@ w4 = mk(64, 7, 7, 8) * 0.1
 fl4 = 2 * (B * 256 * 256) * 64 * (49 * 8)
 cases += [
     ("stem fwd    ", fl4, lambda: e.conv2d_fwd(x4, w4, None, 1, 3, 3, 3, 3, True, 0, 0.2)),
-    ("stem wgrad  ", fl4, lambda: e.conv2d_wgrad(x4, mk(B, 256, 256, 64), 7, 7, 1, 3, 3, True)),
+]
+dy4 = mk(B, 256, 256, 64)
+cases += [
+    ("stem wgrad  ", fl4, lambda: e.conv2d_wgrad(x4, dy4, 7, 7, 1, 3, 3, True)),
 ]
 
 # generator head 7x7 64->3 (padded to 8) @256^2, reflect pad 3 (BN16 tile path)
@@ -92,9 +95,12 @@ bta = torch.rand(256, device=DEV)
 xin = mk(B, 64, 64, 256)
 res = mk(B, 64, 64, 256)
 byts = B * 64 * 64 * 256 * 2
+xin2 = mk(B, 256, 256, 64)
+g64 = torch.rand(64, device=DEV)
+b64 = torch.rand(64, device=DEV)
 cases += [
     ("IN fwd      ", None, lambda: e.instnorm_fwd(xin, g, bta, 1e-3, 1, 0.2, None)),
-    ("IN fwd 256^2", None, lambda: e.instnorm_fwd(mk(B, 256, 256, 64), torch.rand(64, device=DEV), torch.rand(64, device=DEV), 1e-3, 1, 0.2, None)),
+    ("IN fwd 256^2", None, lambda: e.instnorm_fwd(xin2, g64, b64, 1e-3, 1, 0.2, None)),
 ]
 mean_ = torch.rand(B, 256, device=DEV)
 rstd_ = torch.rand(B, 256, device=DEV) + 0.5

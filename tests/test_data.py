@@ -89,3 +89,20 @@ def test_device_prefetcher_cpu_passthrough(local_ctx):
     assert len(batches) == p.train_steps
     ref = list(p.train_epoch(0))
     assert torch.equal(batches[0][0], ref[0][0])
+
+
+def test_folder_images_loading(tmp_path):
+    """--data_dir image-folder loading (PIL decode, RGB, uint8 HWC)."""
+    import numpy as np
+    import PIL.Image
+    from cyclegan_amd.data.pipeline import folder_images
+    d = tmp_path / "trainA"
+    d.mkdir()
+    for i, size in enumerate([(10, 12), (7, 7)]):
+        a = (np.random.rand(size[0], size[1], 3) * 255).astype("uint8")
+        PIL.Image.fromarray(a).save(str(d / f"im{i}.png"))
+    (d / "notes.txt").write_text("ignored")
+    imgs = folder_images(str(d))
+    assert len(imgs) == 2
+    assert imgs[0].shape == (10, 12, 3) and imgs[0].dtype == torch.uint8
+    assert imgs[1].shape == (7, 7, 3)

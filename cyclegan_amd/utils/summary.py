@@ -24,6 +24,10 @@ class Summary:
     def get_writer(self, training: bool):
         return self.writers[0 if training else 1]
 
+    def close(self):
+        for w in self.writers:
+            w.close()
+
     def scalar(self, tag, value, step: int = 0, training: bool = False):
         self.get_writer(training).scalar(tag, float(value), step)
 

@@ -114,6 +114,9 @@ def main(args):
                 utils.plot_cycle(pipe.plot_pairs(), gan, summary, epoch)
             ctx.barrier()
 
+    if summary is not None:
+        summary.close()
+
 
 if __name__ == "__main__":
     parser = argparse.ArgumentParser()

@@ -44,3 +44,29 @@ def test_step_timer():
     s = t.summary()
     assert "a" in s and s["a"] >= 0
     assert t.counts["a"] == 2
+
+
+def test_main_cli_end_to_end(tmp_path):
+    """Reference CLI contract (main.py:405-413): one tiny epoch on CPU
+    writes train+test event files, a checkpoint, and auto-resumes."""
+    out = tmp_path / "run"
+    cmd = [sys.executable, "main.py", "--output_dir", str(out),
+           "--epochs", "1", "--batch_size", "2", "--verbose", "0",
+           "--image_size", "32", "--num_residual_blocks", "1",
+           "--num_train_samples", "4", "--num_test_samples", "2",
+           "--dtype", "fp32"]
+    cwd = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(cmd, cwd=cwd, capture_output=True, text=True,
+                       timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert os.path.exists(out / "checkpoints" / "checkpoint.pt")
+    import glob as g
+    assert g.glob(str(out / "events.out.tfevents.*"))
+    assert g.glob(str(out / "test" / "events.out.tfevents.*"))
+    # auto-resume: second run must report loading the checkpoint
+    r2 = subprocess.run(cmd, cwd=cwd, capture_output=True, text=True,
+                        timeout=600)
+    assert r2.returncode == 0, r2.stderr[-2000:]
+    assert "restor" in (r2.stdout + r2.stderr).lower() or \
+           "resum" in (r2.stdout + r2.stderr).lower() or \
+           "loaded" in (r2.stdout + r2.stderr).lower()

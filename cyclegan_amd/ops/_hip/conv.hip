@@ -1260,8 +1260,10 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   constexpr int API = 16 / NW;        // A glds instructions per wave
   constexpr int DCH = WBN / 8;        // 16-B chunks per D row
   constexpr int DRPI = 64 / DCH;      // D rows per glds instruction
-  constexpr int DPI = (WG_BM / DRPI) / NW;  // D instructions per wave
-  constexpr int NWCW = WBN / 32;      // wave-grid columns
+  constexpr int TDI = (WG_BM * DCH) / 64;   // D glds instructions total
+  constexpr int DPI = TDI >= NW ? TDI / NW : 1;  // per wave (maybe idle)
+  constexpr int NFW = WBN >= 32 ? 2 : 1;    // n-fragments per wave
+  constexpr int NWCW = WBN / (NFW * 16);    // wave-grid columns
   constexpr int KF = (WG_BK / (NW / NWCW)) / 16;  // k-fragments per wave
   extern __shared__ __attribute__((aligned(16))) char wg_smem_raw[];
   auto& sm = *reinterpret_cast<WgSmemT<WBN>*>(wg_smem_raw);
@@ -1302,11 +1304,13 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     a_dkw[j] = tap - a_dkh[j] * p.KW;
   }
   // D: instr i covers DRPI rows; per-wave share = DPI
+  const bool dw_on = w * DPI < TDI;  // waves beyond TDI stage no D
   int d_row[DPI], d_n[DPI];
   #pragma unroll
   for (int j = 0; j < DPI; ++j) {
     int row = (w * DPI + j) * DRPI + lane / DCH;
-    int cb = ((lane % DCH) * 16) ^ (WBN == 128 ? AXOR(row) : DXOR(row));
+    int cb = ((lane % DCH) * 16) ^
+             (WBN == 128 ? AXOR(row) : WBN == 64 ? DXOR(row) : 0);
     d_row[j] = row;
     d_n[j] = n0 + cb / 2;
   }
@@ -1385,19 +1389,21 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
           rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * API + j) * 512],
           16, vo, 0, 0, 0);
     }
-    #pragma unroll
-    for (int j = 0; j < DPI; ++j) {
-      unsigned vo = (d_n[j] < p.Cout && ms + d_row[j] < p.M)
-                        ? dvo[j] : 0xFF000000u;
-      __builtin_amdgcn_raw_ptr_buffer_load_lds(
-          rd, (__attribute__((address_space(3))) void*)&sm.D[buf][(w * DPI + j) * 512],
-          16, vo, 0, 0, 0);
+    if (dw_on) {
+      #pragma unroll
+      for (int j = 0; j < DPI; ++j) {
+        unsigned vo = (d_n[j] < p.Cout && ms + d_row[j] < p.M)
+                          ? dvo[j] : 0xFF000000u;
+        __builtin_amdgcn_raw_ptr_buffer_load_lds(
+            rd, (__attribute__((address_space(3))) void*)&sm.D[buf][(w * DPI + j) * 512],
+            16, vo, 0, 0, 0);
+      }
     }
   };
 
-  v4f acc[KF][2] = {};
+  v4f acc[KF][NFW] = {};
   const int wr = w / NWCW, wc = w % NWCW;
-  const int wk0 = wr * (KF * 16), wn0 = wc * 32;
+  const int wk0 = wr * (KF * 16), wn0 = wc * (NFW * 16);
   const int fr = lane & 15, fg = lane >> 4;
   const int jg = lane & 15;
   // tr-read per-lane address components (bytes)
@@ -1428,23 +1434,26 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
         v4bfx hi = tr16_read(Ab + row * 256 + (cbl ^ AXOR(row)));
         a[kf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
-      v8bf bfr[2];
+      v8bf bfr[NFW];
       #pragma unroll
-      for (int nf = 0; nf < 2; ++nf) {
+      for (int nf = 0; nf < NFW; ++nf) {
         int cbl = (wn0 + nf * 16) * 2 + tr_cb_a;
         int row = kk + fg * 8 + tr_row_a;
         v4bfx lo = tr16_read(Db + row * (WBN * 2) +
-                             (cbl ^ (WBN == 128 ? AXOR(row) : DXOR(row))));
+                             (cbl ^ (WBN == 128 ? AXOR(row)
+                                     : WBN == 64 ? DXOR(row) : 0)));
         row += 4;
         v4bfx hi = tr16_read(Db + row * (WBN * 2) +
-                             (cbl ^ (WBN == 128 ? AXOR(row) : DXOR(row))));
+                             (cbl ^ (WBN == 128 ? AXOR(row)
+                                     : WBN == 64 ? DXOR(row) : 0)));
         bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
       #pragma unroll
-      for (int kf = 0; kf < KF; ++kf) {
-        acc[kf][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[kf], bfr[0], acc[kf][0], 0, 0, 0);
-        acc[kf][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[kf], bfr[1], acc[kf][1], 0, 0, 0);
-      }
+      for (int kf = 0; kf < KF; ++kf)
+        #pragma unroll
+        for (int nf = 0; nf < NFW; ++nf)
+          acc[kf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[kf], bfr[nf], acc[kf][nf], 0, 0, 0);
     }
     __syncthreads();
   }
@@ -1454,8 +1463,9 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     long chunk = (((long)sl * p.ktiles + kt) * p.ntiles + nt) *
                  ((long)WBN * WG_BK);
     #pragma unroll
-    for (int nf = 0; nf < 2; ++nf) {
+    for (int nf = 0; nf < NFW; ++nf) {
       int nl = wn0 + nf * 16 + fr;
+      if (nl >= WBN) continue;
       #pragma unroll
       for (int kf = 0; kf < KF; ++kf) {
         int kl = wk0 + kf * 16 + fg * 4;
@@ -1853,9 +1863,10 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
     p.dw = (float*)dw.mutable_data_ptr();
     int wbn = (conv_nw() == 8 && conv_bn() == 128 && (p.Cout % 128) == 0)
                   ? 128 : WG_BN;
-    if (wbn == 128) {
-      p.ntiles = (p.Cout + 127) / 128;
-      // re-balance the split-M slices for the halved tile count
+    if (p.Cout <= 16) wbn = 16;  // tiny-N heads: 16-wide n-tiles
+    if (wbn != WG_BN) {
+      p.ntiles = (p.Cout + wbn - 1) / wbn;
+      // re-balance the split-M slices for the changed tile count
       long mchunks = (p.M + WG_BM - 1) / WG_BM;
       static int tgt2 = []() {
         const char* e = getenv("CYG_WG_BLOCKS");
@@ -1881,6 +1892,9 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
       (void)init;
       hipLaunchKernelGGL((wgrad_glds_kernel<8, 128>), grid, dim3(512), SMB,
                          stream, p);
+    } else if (wbn == 16) {
+      hipLaunchKernelGGL((wgrad_glds_kernel<8, 16>), grid, dim3(512),
+                         sizeof(WgSmemT<16>), stream, p);
     } else if (conv_nw() == 8) {
       hipLaunchKernelGGL((wgrad_glds_kernel<8, 64>), grid, dim3(512),
                          sizeof(WgSmemT<64>), stream, p);

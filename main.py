@@ -63,8 +63,9 @@ def main(args):
         os.makedirs(args.output_dir, exist_ok=True)
     ctx.barrier()
 
-    np.random.seed(1234)
-    torch.manual_seed(1234)
+    # reference seeds np/tf with 1234 (main.py:366-367); --seed overrides
+    np.random.seed(args.seed)
+    torch.manual_seed(args.seed)
 
     args.global_batch_size = ctx.world_size * args.batch_size
     if args.dtype:

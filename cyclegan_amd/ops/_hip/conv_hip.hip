@@ -1658,9 +1658,9 @@ static void launch_glds_s(const ConvParams& p, dim3 grid, hipStream_t stream) {
     q.ntiles = (p.Cout + 15) / 16;
     dim3 g2((long)q.mtiles * q.ntiles);
     switch (p.stride) {
-      case 1: launch_one_glds<IS_CONVT, 1, 2, 4, 16>(q, g2, stream); return;
-      case 2: launch_one_glds<IS_CONVT, 2, 2, 4, 16>(q, g2, stream); return;
-      default: launch_one_glds<IS_CONVT, 0, 2, 4, 16>(q, g2, stream); return;
+      case 1: launch_one_glds<IS_CONVT, 1, 2, 8, 16>(q, g2, stream); return;
+      case 2: launch_one_glds<IS_CONVT, 2, 2, 8, 16>(q, g2, stream); return;
+      default: launch_one_glds<IS_CONVT, 0, 2, 8, 16>(q, g2, stream); return;
     }
   }
   if (conv_nw() == 8 && conv_bn() == 128 && (p.Cout % 128) == 0) {

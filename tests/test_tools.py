@@ -70,3 +70,41 @@ def test_main_cli_end_to_end(tmp_path):
     assert "restor" in (r2.stdout + r2.stderr).lower() or \
            "resum" in (r2.stdout + r2.stderr).lower() or \
            "loaded" in (r2.stdout + r2.stderr).lower()
+
+
+def test_bench_json_contract(tmp_path):
+    """The driver parses bench.py's final line as JSON with a fixed
+    schema; guard every required key and the value semantics."""
+    import json
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--output_dir", str(tmp_path)],
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    d = json.loads(r.stdout.strip().splitlines()[-1])
+    for k in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+              "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+              "dtype", "data", "config"):
+        assert k in d, k
+    assert d["metric"].startswith("images/sec")
+    assert d["higher_is_better"] is True and d["scaling"] == "weak"
+    assert d["data"] == "synthetic" and d["n_gpus"] == 1
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["config"]["global_batch"] == d["n_gpus"] * d["config"]["per_gpu_batch"]
+
+
+def test_fused_adam_graph_mode_lr_math(tmp_path):
+    """advance_lr must track the eager TF bias-correction schedule."""
+    import math
+    from cyclegan_amd.ops.adam import FusedAdam
+    p = torch.zeros(4)
+    g = torch.zeros(4)
+    opt = FusedAdam(p, g)
+    opt.prepare_graph()
+    assert opt.graph_mode and opt._lr_t_dev is not None
+    for t in range(1, 5):
+        opt.advance_lr()
+        want = opt.lr * math.sqrt(1 - opt.b2 ** t) / (1 - opt.b1 ** t)
+        assert opt.t == t
+        assert abs(opt._lr_t_dev.item() - want) < 1e-9 * (1 + abs(want))

@@ -1,6 +1,9 @@
 """Minimal TensorBoard event-file writer (no tensorboard package needed).
 
-Writes standard TFRecord-framed Event protobufs that TensorBoard reads:
+Backs the reference's two-writer Summary layout (reference utils.py:21-24;
+tags written via utils.py:29-37): one events.out.tfevents.* file per
+logdir. Writes standard TFRecord-framed Event protobufs that TensorBoard
+reads:
 scalars and (PNG-encoded) images. Protobuf messages are hand-encoded on the
 wire (the schema is tiny and frozen):
 

@@ -8,8 +8,9 @@ on the MI355X execution model:
 - losses pre-scaled by 1/global_batch_size so SUM all-reduce over DP
   replicas is the exact global mean (main.py:172-174);
 - per-step: one shared forward mega-graph; 4 gradient passes restricted to
-  each model's variables (``torch.autograd.backward(..., inputs=group)`` ==
-  TF ``optimizer.minimize(var_list=...)``), each followed immediately by an
+  each model's variables (``torch.autograd.grad(loss, group.params)`` ==
+  TF ``optimizer.minimize(var_list=...)``; grads land in the flat buffer
+  via one batched multi-tensor copy), each followed immediately by an
   async RCCL all-reduce of that group's flat gradient, overlapping the next
   backward (reference runs the 4 groups serially, main.py:249-260);
 - G's adversarial gradient flows through the frozen discriminator (the

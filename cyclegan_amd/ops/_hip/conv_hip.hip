@@ -14,10 +14,16 @@
 // (fixed n, k contiguous) are vector loads AND single ds_read_b128 MFMA
 // fragments.
 //
-// Tiling: 128x64 block tile, BK=64, 4 waves (2x2), 64x32 per wave,
-// mfma_f32_16x16x32_bf16, fragments: A row = lane&15, k = (lane>>4)*8..+8;
+// Tiling (conv_glds_kernel, the production path): BM=128 rows, BK=64,
+// n-tile width 128/64/16 by Cout (8 waves for 128/64, Cout<=16 heads get
+// 16-wide NF=1 tiles), glds (buffer_load_lds) double-buffered staging
+// with the XOR bank swizzle riding the gather address, incremental
+// im2col addressing, XCD-chunked blockIdx->tile mapping;
+// mfma_f32_16x16x32_bf16 fragments: A row = lane&15, k = (lane>>4)*8..+8;
 // D col = lane&15, row = (lane>>4)*4 + reg (verified by tests/test_ops_gpu.py
-// via the mfma_probe binding and oracle comparisons).
+// via the mfma_probe binding and oracle comparisons). A halo-tiled direct
+// head-conv kernel exists behind CYG_HALO=1 (see NOTES.md). The plain
+// conv_gemm_kernel remains as the unaligned-channel fallback.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>

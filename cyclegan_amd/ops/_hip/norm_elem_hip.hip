@@ -36,7 +36,7 @@ __global__ __launch_bounds__(NT) void in_reduce_kernel(
   int rstep = NT / gpr;
   int rof = tid / gpr;
   float s[8] = {}, q[8] = {};
-  #pragma unroll 2
+  #pragma unroll 4
   for (long r = r0 + rof; r < r1; r += rstep) {
     v8s v = *(const v8s*)(xb + r * C + g * 8);
     #pragma unroll
@@ -172,7 +172,7 @@ __global__ __launch_bounds__(NT) void in_norm_kernel(
     int g = tid % gpr;
     int rstep = NT / gpr;
     int rof = tid / gpr;
-    #pragma unroll 2
+    #pragma unroll 4
     for (long r = r0 + rof; r < r1; r += rstep) {
       long off = base + r * C + g * 8;
       v8s v = *(const v8s*)(x + off);
@@ -223,7 +223,7 @@ __global__ __launch_bounds__(NT) void in_bwd_reduce_kernel(
     mv[j] = mean[(long)b * C + g * 8 + j];
     rv[j] = rstd[(long)b * C + g * 8 + j];
   }
-  #pragma unroll 2
+  #pragma unroll 4
   for (long r = r0 + rof; r < r1; r += rstep) {
     long off = base + r * C + g * 8;
     v8s dv = *(const v8s*)(dy + off);
@@ -356,7 +356,7 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
     int g = tid % gpr;
     int rstep = NT / gpr;
     int rof = tid / gpr;
-    #pragma unroll 2
+    #pragma unroll 4
     for (long r = r0 + rof; r < r1; r += rstep) {
       long off = base + r * C + g * 8;
       v8s dv = *(const v8s*)(dy + off);

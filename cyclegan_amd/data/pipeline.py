@@ -76,8 +76,10 @@ def folder_images(path: str) -> List[torch.Tensor]:
     for name in sorted(os.listdir(path)):
         if name.lower().endswith((".jpg", ".jpeg", ".png")):
             with PIL.Image.open(os.path.join(path, name)) as im:
+                # np.asarray of a PIL image is read-only; copy so the
+                # tensor owns writable memory (torch warns otherwise)
                 out.append(torch.from_numpy(
-                    np.asarray(im.convert("RGB"), dtype="uint8")))
+                    np.array(im.convert("RGB"), dtype="uint8")))
     return out
 
 

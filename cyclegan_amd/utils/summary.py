@@ -17,6 +17,16 @@ from .tb_writer import EventWriter
 class Summary:
     def __init__(self, output_dir: str):
         self.dpi = 120
+        # reference sets plt.style.use('seaborn-deep') at import
+        # (utils.py:19); modern matplotlib renamed the style
+        try:
+            import matplotlib.pyplot as plt
+            for name in ("seaborn-deep", "seaborn-v0_8-deep"):
+                if name in plt.style.available:
+                    plt.style.use(name)
+                    break
+        except Exception:
+            pass
         import os
         self.writers = [EventWriter(output_dir),
                         EventWriter(os.path.join(output_dir, "test"))]

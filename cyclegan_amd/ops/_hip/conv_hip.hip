@@ -2043,6 +2043,38 @@ at::Tensor tr_probe(int64_t mode) {
 }
 
 // ---- MFMA layout probe (test aid): C[16,16] = A[16,32] @ Bt[16,32]^T ----
+// MX-scaled fp8 MFMA probe (round-2 prep): raw per-lane operand dump for
+// mapping the 32x32x64 f8f6f4 fragment layout empirically, exactly how
+// mfma_probe established the bf16 16x16x32 layout. Operands are the raw
+// 32 bytes each lane supplies (a/b as [64][8] int32); scale operands use
+// cbsz/blgp = 0 (e4m3 A and B) and a single scale byte broadcast.
+typedef int v8i_ __attribute__((ext_vector_type(8)));
+typedef float v16f_ __attribute__((ext_vector_type(16)));
+__global__ void mx_probe_kernel(const int* a, const int* b, float* d,
+                                int sa, int sb) {
+  int lane = threadIdx.x & 63;
+  v8i_ av = *(const v8i_*)(a + lane * 8);
+  v8i_ bv = *(const v8i_*)(b + lane * 8);
+  v16f_ acc = {};
+  acc = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+      av, bv, acc, 0, 0, 0, sa, 0, sb);
+  #pragma unroll
+  for (int r = 0; r < 16; ++r) d[lane * 16 + r] = acc[r];
+}
+
+at::Tensor mx_probe(at::Tensor a, at::Tensor b, int64_t sa, int64_t sb) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kInt &&
+              a.sizes() == at::IntArrayRef({64, 8}) &&
+              b.sizes() == at::IntArrayRef({64, 8}));
+  auto d = at::empty({64, 16}, a.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(mx_probe_kernel, dim3(1), dim3(64), 0,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     (const int*)a.contiguous().const_data_ptr(),
+                     (const int*)b.contiguous().const_data_ptr(),
+                     (float*)d.mutable_data_ptr(), (int)sa, (int)sb);
+  return d;
+}
+
 __global__ void mfma_probe_kernel(const short* a, const short* bt, float* c) {
   int lane = threadIdx.x & 63;
   int fr = lane & 15, fg = lane >> 4;

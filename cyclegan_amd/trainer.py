@@ -93,9 +93,40 @@ class CycleGAN:
     def _cast(self, t: torch.Tensor) -> torch.Tensor:
         return t.to(self.device, self.compute_dtype, non_blocking=True)
 
+    _TRAIN_KEYS = ("loss_G/loss", "loss_G/cycle", "loss_G/identity",
+                   "loss_G/total", "loss_F/loss", "loss_F/cycle",
+                   "loss_F/identity", "loss_F/total", "loss_X/loss",
+                   "loss_Y/loss")
+    _TEST_KEYS = _TRAIN_KEYS + ("error/MAE(X, F(G(X)))",
+                                "error/MAE(Y, G(F(Y)))",
+                                "error/MAE(X, F(X))", "error/MAE(Y, G(Y))")
+
+    def _zero_losses(self, keys) -> Dict[str, torch.Tensor]:
+        z = torch.zeros((), device=self.device)
+        return {k: z for k in keys}
+
+    def _empty_train_step(self) -> Dict[str, torch.Tensor]:
+        """Short final global batch under DP: this rank's slice is empty
+        (the reference's MirroredStrategy replicas idle the same way,
+        main.py:80-81). Contribute zero gradients but still join the
+        all-reduces and take the identical optimizer step — after the
+        SUM all-reduce every rank applies the same update, so replicas
+        stay in sync."""
+        for g in self.groups.values():
+            g.flat_grad.zero_()
+            self.sync.launch(g.flat_grad)
+        self.sync.wait_all()
+        for opt in self.optimizers.values():
+            opt.step()
+        for g in self.groups.values():
+            g.bump_versions()
+        return self._zero_losses(self._TRAIN_KEYS)
+
     def train_step(self, x, y) -> Dict[str, torch.Tensor]:
         x, y = self._cast(x), self._cast(y)
         b = x.shape[0]
+        if b == 0:
+            return self._empty_train_step()
         # batched generator calls: every op is per-sample (convs, per-sample
         # InstanceNorm stats, per-sample losses), so G(cat(x,y)) is
         # numerically identical to G(x), G(y) — fewer, larger kernels.
@@ -167,6 +198,8 @@ class CycleGAN:
     @torch.no_grad()
     def test_step(self, x, y) -> Dict[str, torch.Tensor]:
         x, y = self._cast(x), self._cast(y)
+        if x.shape[0] == 0:  # empty DP slice of a short final batch
+            return self._zero_losses(self._TEST_KEYS)
         fake_x, fake_y, cycle_x, cycle_y = self.cycle_step(x, y)
 
         discriminate_fake_x = self.X(fake_x)

@@ -80,3 +80,48 @@ def test_two_rank_matches_single_rank(tmp_path):
     # (the 2-rank run reports the SUM-all-reduced value only at epoch end; here
     #  we just check single-rank loss is finite and comparable in magnitude)
     assert abs(r["loss_G/total"].item()) < 100
+
+
+def _worker_short_batch(rank, world, port, outdir):
+    """Final global batch of 1 sample: rank 0 gets it, rank 1's slice is
+    empty — both must finish the epoch with IDENTICAL replicas."""
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    from cyclegan_amd.parallel import DistContext
+    torch.manual_seed(1234)
+    ctx = DistContext(device=torch.device("cpu"))
+    gan = CycleGAN(_make_args(outdir, 1, 2), ctx)
+    x, y = _data()
+    # step 1: full batch (1 sample per rank); step 2: short (rank 1 empty)
+    gan.train_step(x[rank:rank + 1], y[rank:rank + 1])
+    if rank == 0:
+        r = gan.train_step(x[1:2], y[1:2])
+    else:
+        r = gan.train_step(x[:0], y[:0])
+    for v in r.values():
+        assert torch.isfinite(v)
+    te = gan.test_step(x[:0], y[:0]) if rank == 1 else gan.test_step(x[:1], y[:1])
+    assert set(te) == set(gan._TEST_KEYS)
+    torch.save({"flat_G": gan.groups["G"].flat_param.detach().clone(),
+                "t": gan.optimizers["G"].t},
+               os.path.join(outdir, f"short_rank{rank}.pt"))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_short_final_batch_keeps_replicas_synced(tmp_path):
+    port = 29517
+    ctxq = mp.get_context("spawn")
+    procs = [ctxq.Process(target=_worker_short_batch,
+                          args=(r, 2, port, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    r0 = torch.load(os.path.join(str(tmp_path), "short_rank0.pt"))
+    r1 = torch.load(os.path.join(str(tmp_path), "short_rank1.pt"))
+    assert r0["t"] == r1["t"] == 2
+    assert torch.equal(r0["flat_G"], r1["flat_G"])

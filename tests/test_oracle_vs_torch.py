@@ -81,3 +81,25 @@ def test_conv_transpose2d_cpu_matches_torch(b, cin, cout, k):
     assert y.shape == ref.shape
     assert torch.allclose(y, ref, atol=1e-4, rtol=1e-4), \
         (y - ref).abs().max()
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.integers(1, 3), st.integers(1, 6), st.integers(2, 9),
+       st.booleans())
+def test_instance_norm_cpu_matches_torch(b, c, hw, with_residual):
+    """IN oracle vs torch.nn.functional.instance_norm at tfa's eps=1e-3
+    (per-(sample,channel) stats, affine)."""
+    g = torch.Generator().manual_seed(b * 31 + c * 7 + hw)
+    x = torch.randn(b, hw, hw, c, generator=g)
+    gamma = torch.randn(c, generator=g) * 0.02
+    beta = torch.randn(c, generator=g) * 0.1
+    res = torch.randn(b, hw, hw, c, generator=g) if with_residual else None
+    y = ops.instance_norm(x, gamma, beta, eps=1e-3, act="relu",
+                          residual=res)
+    ref = F.instance_norm(_nchw(x), weight=gamma, bias=beta,
+                          eps=1e-3).permute(0, 2, 3, 1)
+    if res is not None:
+        ref = ref + res
+    ref = torch.relu(ref)
+    assert torch.allclose(y, ref, atol=1e-5, rtol=1e-4), \
+        (y - ref).abs().max()

@@ -29,8 +29,8 @@ from cyclegan_amd.trainer import CycleGAN
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
-    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--steps", type=int, default=100)
+    ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--batch_size", type=int, default=4, help="per-GPU batch")
     ap.add_argument("--image_size", type=int, default=256)
     ap.add_argument("--num_residual_blocks", type=int, default=9)
@@ -47,6 +47,8 @@ def main():
         args.num_residual_blocks = 1
         args.batch_size = 1
         args.dtype = "fp32"
+        args.steps = min(args.steps, 20)
+        args.warmup = min(args.warmup, 5)
 
     torch.manual_seed(1234)
     args.global_batch_size = ctx.world_size * args.batch_size

@@ -129,7 +129,7 @@ class Pipeline:
         self.train_steps = math.ceil(self.num_train / self.global_batch)
         self.test_steps = math.ceil(self.num_test / self.global_batch)
 
-        ih = (287 * image_size) // 256  # scale the 286/256 ratio with size
+        ih = (286 * image_size) // 256  # scale the 286/256 ratio with size
         ishape, cshape = (ih, ih), (image_size, image_size)
         # .map().cache(): augmentation frozen per-sample for the run
         self.trainA = [preprocess_train(im, gen, ishape, cshape)

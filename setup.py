@@ -16,8 +16,11 @@ from torch.utils import cpp_extension  # noqa: E402
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 HIP_DIR = os.path.join(HERE, "cyclegan_amd", "ops", "_hip")
-sources = sorted(glob.glob(os.path.join(HIP_DIR, "*.hip")) +
-                 glob.glob(os.path.join(HIP_DIR, "*.cpp")))
+# Exclude torch-hipify build artifacts (*_hip.hip) — they are generated
+# copies of the hand-written sources and must not be compiled or tracked.
+sources = sorted(s for s in (glob.glob(os.path.join(HIP_DIR, "*.hip")) +
+                             glob.glob(os.path.join(HIP_DIR, "*.cpp")))
+                 if not s.endswith("_hip.hip"))
 
 ext_modules = []
 if sources:

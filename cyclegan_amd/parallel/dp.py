@@ -30,11 +30,19 @@ class DistContext:
         self.rank = int(os.environ.get("RANK", "0"))
         self.world_size = int(os.environ.get("WORLD_SIZE", "1"))
         self.local_rank = int(os.environ.get("LOCAL_RANK", str(self.rank)))
-        self.distributed = self.world_size > 1
+        # CYG_FORCE_DIST=1: create the RCCL communicator even at
+        # world_size 1 (on-silicon shakeout of the ProcessGroupNCCL path
+        # within a 1-GPU lease; RCCL refuses two ranks on one device —
+        # "Duplicate GPU detected", see profiles/rccl_shakeout.md)
+        self.distributed = (self.world_size > 1 or
+                            os.environ.get("CYG_FORCE_DIST") == "1")
         if device is not None:
             self.device = device
         elif torch.cuda.is_available():
-            self.device = torch.device("cuda", self.local_rank)
+            # modulo: lets an N-rank job oversubscribe fewer GPUs (RCCL
+            # shakeout of the multi-rank path on a 1-GPU box)
+            self.device = torch.device(
+                "cuda", self.local_rank % torch.cuda.device_count())
         else:
             self.device = torch.device("cpu")
         if self.device.type == "cuda":

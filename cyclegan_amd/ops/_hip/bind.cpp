@@ -39,6 +39,7 @@ std::vector<at::Tensor> persample_loss_bwd(at::Tensor, at::Tensor, at::Tensor,
 at::Tensor persample_loss_const_bwd(at::Tensor, double, at::Tensor, bool);
 void adam_step_dev(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                    double, double, double);
+void shadow_gather(at::Tensor, at::Tensor, at::Tensor);
 void adam_step(at::Tensor, at::Tensor, at::Tensor, at::Tensor, double, double,
                double, double, int64_t);
 }  // namespace cyg
@@ -67,4 +68,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("persample_loss_const_bwd", &cyg::persample_loss_const_bwd);
   m.def("adam_step", &cyg::adam_step);
   m.def("adam_step_dev", &cyg::adam_step_dev);
+  m.def("shadow_gather", &cyg::shadow_gather);
 }

@@ -34,7 +34,7 @@ import torch.nn.functional as F
 
 from . import backend
 from .shadow import (compute_weight, compute_weight_t, compute_weight_p,
-                     compute_weight_tp, fp8_weight_shadow)
+                     compute_weight_tp, compute_bias_p, fp8_weight_shadow)
 
 ACT_NONE, ACT_RELU, ACT_LRELU, ACT_TANH = 0, 1, 2, 3
 
@@ -116,9 +116,7 @@ class _ConvFn(torch.autograd.Function):
         cout = w.shape[0]
         xp = _pad_channels(x).contiguous()
         wc = compute_weight_p(w, x)
-        bc = compute_weight(bias, x) if bias is not None else None
-        if bc is not None and cout < 8:
-            bc = _pad_channels(bc.view(1, 1, 1, -1)).view(-1)
+        bc = compute_bias_p(bias, x) if bias is not None else None
         y = ext.conv2d_fwd(xp, wc, bc, stride, *pads, reflect, act, slope)
         if cout < 8:
             y = y[..., :cout].contiguous()
@@ -184,9 +182,7 @@ class _ConvFp8Fn(torch.autograd.Function):
         sx = (448.0 / amax).clamp(max=65504.0)
         xq = ext.quant_fp8(xp, sx)
         dq = (sx * sw).reciprocal()
-        bc = compute_weight(bias, x) if bias is not None else None
-        if bc is not None and cout < 8:
-            bc = _pad_channels(bc.view(1, 1, 1, -1)).view(-1)
+        bc = compute_bias_p(bias, x) if bias is not None else None
         y = ext.conv2d_fp8_fwd(xq, wq, dq, bc, stride, *pads, reflect, act, slope)
         if cout < 8:
             y = y[..., :cout].contiguous()

@@ -19,6 +19,23 @@ _cache_p: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
 _cache_tp: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
 _cache_f8: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
 
+# arena-backed forms (ops.arena): master param -> {form: bf16 view}.
+# Views stay fresh because the trainer refreshes the arena after every
+# parameter mutation (optimizer step / checkpoint load / broadcast).
+_arena_forms: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
+
+
+def register_arena(arena):
+    for p, forms in arena._by_param.items():
+        _arena_forms[p] = forms
+
+
+def _arena(p: torch.Tensor, form: str, like: torch.Tensor):
+    if like.dtype != torch.bfloat16:
+        return None
+    forms = _arena_forms.get(p)
+    return None if forms is None else forms.get(form)
+
 
 def bf16_shadow(t: torch.Tensor) -> torch.Tensor:
     """Return a cached bf16 copy of ``t`` (refreshed when t changes in-place)."""
@@ -35,11 +52,25 @@ def bf16_shadow(t: torch.Tensor) -> torch.Tensor:
 
 def compute_weight(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
     """Weight in the compute dtype of activation ``like`` (no autograd edge)."""
+    v = _arena(w, "plain", like)
+    if v is not None:
+        return v
     if like.dtype == torch.bfloat16 and w.dtype != torch.bfloat16:
         return bf16_shadow(w)
     if w.dtype != like.dtype:
         return w.detach().to(like.dtype)
     return w.detach()
+
+
+def compute_bias_p(b: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
+    """Bias padded to >= 8 channels in the compute dtype."""
+    v = _arena(b, "bias_p", like)
+    if v is not None:
+        return v
+    bc = compute_weight(b, like)
+    if bc.numel() < 8:
+        bc = torch.nn.functional.pad(bc, (0, 8 - bc.numel()))
+    return bc
 
 
 def _pad_dims(w: torch.Tensor) -> torch.Tensor:
@@ -56,6 +87,9 @@ def _pad_dims(w: torch.Tensor) -> torch.Tensor:
 
 def compute_weight_p(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
     """Padded (dims >= 8) compute-dtype weight, cached per master version."""
+    v = _arena(w, "p", like)
+    if v is not None:
+        return v
     ent = _cache_p.get(w)
     ver = w._version
     key = (ver, like.dtype, "p")
@@ -73,6 +107,9 @@ def compute_weight_p(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
 
 def compute_weight_tp(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
     """Padded channel-transposed weight ([I,kh,kw,O], dims >= 8)."""
+    v = _arena(w, "tp", like)
+    if v is not None:
+        return v
     ent = _cache_tp.get(w)
     ver = w._version
     key = (ver, like.dtype, "tp")
@@ -92,6 +129,9 @@ def compute_weight_t(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
     compute dtype — the B^T operand of the adjoint (dgrad) kernels. Cached
     per master-version like the bf16 shadow (the backward of each of the
     generator's 3 calls per step reuses it)."""
+    v = _arena(w, "t", like)
+    if v is not None:
+        return v
     ent = _cache_t.get(w)
     ver = w._version
     if ent is not None and ent[0] == ver and ent[1] == like.dtype:

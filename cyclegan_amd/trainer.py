@@ -28,7 +28,7 @@ from typing import Dict, Optional
 import torch
 
 from .models import Generator, Discriminator
-from .ops import MAE, MSE, MSE_const
+from .ops import MAE, MSE, MSE_const, backend
 from .ops.adam import FusedAdam
 from .parallel import DistContext, GradSync, FlatParamGroup
 
@@ -68,6 +68,24 @@ class CycleGAN:
         self.optimizers = {name: FusedAdam(g.flat_param, g.flat_grad)
                            for name, g in self.groups.items()}
         self.sync = GradSync(ctx, timing=getattr(args, "verbose", 1) == 2)
+
+        # shadow arenas: every bf16 compute form of a group refreshed by
+        # ONE gather kernel after each optimizer step (ops/arena.py)
+        self.arenas = {}
+        if (self.device.type == "cuda"
+                and self.compute_dtype == torch.bfloat16
+                and backend.load_ext() is not None):
+            from .ops.arena import ShadowArena
+            from .ops import shadow as _shadow
+            for name, m in (("G", self.G), ("F", self.F),
+                            ("X", self.X), ("Y", self.Y)):
+                a = ShadowArena(self.groups[name], m)
+                _shadow.register_arena(a)
+                self.arenas[name] = a
+
+    def _refresh_shadows(self):
+        for a in self.arenas.values():
+            a.refresh()
 
     # ---- loss functions (reference main.py:172-195) ----
 
@@ -120,6 +138,7 @@ class CycleGAN:
             opt.step()
         for g in self.groups.values():
             g.bump_versions()
+        self._refresh_shadows()
         return self._zero_losses(self._TRAIN_KEYS)
 
     def train_step(self, x, y) -> Dict[str, torch.Tensor]:
@@ -130,43 +149,57 @@ class CycleGAN:
         # batched generator calls: every op is per-sample (convs, per-sample
         # InstanceNorm stats, per-sample losses), so G(cat(x,y)) is
         # numerically identical to G(x), G(y) — fewer, larger kernels.
+        #
+        # The cycle inputs are detached: under the reference's var_list
+        # restriction (main.py:249-260) no optimizer ever follows the
+        # fake_y -> G path out of F's cycle branch (F_cycle is only in
+        # F_total, whose grads stop at F's own weights), so cutting it is
+        # gradient-identical and lets F run as ONE 3b-batched call without
+        # autograd descending a numerically-zero path into G.
         g_out = self.G(torch.cat([x, y]))     # -> fake_y, same_y
         fake_y, same_y = g_out[:b], g_out[b:]
-        f_out = self.F(torch.cat([y, x]))     # -> fake_x, same_x
-        fake_x, same_x = f_out[:b], f_out[b:]
+        f_out = self.F(torch.cat([y, x, fake_y.detach()]))
+        fake_x, same_x, cycle_x = f_out[:b], f_out[b:2 * b], f_out[2 * b:]
+        cycle_y = self.G(fake_x.detach())
 
+        # ONE discriminator pass per fake, shared by the adversarial and
+        # discriminator losses (the reference recomputes X(fake_x) at
+        # main.py:239-245 with identical values; TF graph-mode CSE merges
+        # them the same way). X's own update never reaches fake_x: the
+        # path stops at X's first conv weights.
         discriminate_fake_x = self.X(fake_x)
         discriminate_fake_y = self.Y(fake_y)
 
         G_loss = self.generator_loss(discriminate_fake_y)
         F_loss = self.generator_loss(discriminate_fake_x)
-        G_cycle_loss = self.cycle_loss(y, self.G(fake_x))
-        F_cycle_loss = self.cycle_loss(x, self.F(fake_y))
+        G_cycle_loss = self.cycle_loss(y, cycle_y)
+        F_cycle_loss = self.cycle_loss(x, cycle_x)
         G_identity_loss = self.identity_loss(y, same_y)
         F_identity_loss = self.identity_loss(x, same_x)
         G_total = G_loss + G_cycle_loss + G_identity_loss
         F_total = F_loss + F_cycle_loss + F_identity_loss
 
-        # discriminator pass on real + re-discriminated (detached) fakes,
-        # batched (again identical per-sample math)
-        dx = self.X(torch.cat([x, fake_x.detach()]))
-        dy_ = self.Y(torch.cat([y, fake_y.detach()]))
-        X_loss = self.discriminator_loss(dx[:b], dx[b:])
-        Y_loss = self.discriminator_loss(dy_[:b], dy_[b:])
+        X_loss = self.discriminator_loss(self.X(x), discriminate_fake_x)
+        Y_loss = self.discriminator_loss(self.Y(y), discriminate_fake_y)
 
         # torch.autograd.grad (not .backward): grads come back as fresh
         # tensors and land in the flat buffer via ONE batched multi-tensor
         # copy per group, skipping AccumulateGrad's per-param add into the
         # .grad views (~300 tiny launches/step) and the flat zero-fill.
+        # retain_graph through X's pass: the shared discriminate_fake_*
+        # subgraphs are traversed by both the generator and discriminator
+        # passes.
         self.groups["G"].set_grads(
             torch.autograd.grad(G_total, self.groups["G"].params,
                                 retain_graph=True))
         self.sync.launch(self.groups["G"].flat_grad)
         self.groups["F"].set_grads(
-            torch.autograd.grad(F_total, self.groups["F"].params))
+            torch.autograd.grad(F_total, self.groups["F"].params,
+                                retain_graph=True))
         self.sync.launch(self.groups["F"].flat_grad)
         self.groups["X"].set_grads(
-            torch.autograd.grad(X_loss, self.groups["X"].params))
+            torch.autograd.grad(X_loss, self.groups["X"].params,
+                                retain_graph=True))
         self.sync.launch(self.groups["X"].flat_grad)
         self.groups["Y"].set_grads(
             torch.autograd.grad(Y_loss, self.groups["Y"].params))
@@ -177,6 +210,7 @@ class CycleGAN:
             opt.step()
         for g in self.groups.values():
             g.bump_versions()  # invalidate bf16 shadow caches (see flat.py)
+        self._refresh_shadows()
 
         return {
             "loss_G/loss": G_loss.detach(), "loss_G/cycle": G_cycle_loss.detach(),
@@ -259,6 +293,7 @@ class CycleGAN:
         self.Y.load_state_dict(state["Y"])
         for name in ("G", "F", "X", "Y"):
             self.optimizers[name].load_state_dict(state[f"{name}_optimizer"])
+        self._refresh_shadows()
         print(f"\nloaded checkpoint from {self.checkpoint_path}\n")
         return True
 

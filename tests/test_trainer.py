@@ -121,3 +121,43 @@ def test_loss_scaling_by_global_batch(tmp_path, ctx):
     gan.global_batch_size = 2
     r2 = gan.test_step(x, y)
     assert abs(r["loss_G/total"].item() * 2 - r2["loss_G/total"].item()) < 1e-5
+
+
+def test_train_step_grads_match_reference_structure(tmp_path, ctx):
+    """The shared-discriminator-node / batched-F train_step must produce
+    the same gradients as the reference's explicit-recompute structure
+    (/root/reference/main.py:207-262: separate F calls, re-discriminated
+    detached fakes)."""
+    torch.manual_seed(3)
+    args = make_args(tmp_path)
+    gan = CycleGAN(args, ctx)
+    x = torch.rand(2, 32, 32, 3) * 2 - 1
+    y = torch.rand(2, 32, 32, 3) * 2 - 1
+
+    # --- reference-structure gradients ---
+    fake_y = gan.G(x)
+    same_y = gan.G(y)
+    fake_x = gan.F(y)
+    same_x = gan.F(x)
+    G_total = (gan.generator_loss(gan.Y(fake_y))
+               + gan.cycle_loss(y, gan.G(fake_x))
+               + gan.identity_loss(y, same_y))
+    F_total = (gan.generator_loss(gan.X(fake_x))
+               + gan.cycle_loss(x, gan.F(fake_y))
+               + gan.identity_loss(x, same_x))
+    dx = gan.X(torch.cat([x, fake_x.detach()]))
+    dy_ = gan.Y(torch.cat([y, fake_y.detach()]))
+    X_loss = gan.discriminator_loss(dx[:2], dx[2:])
+    Y_loss = gan.discriminator_loss(dy_[:2], dy_[2:])
+    want = {}
+    for name, loss in (("G", G_total), ("F", F_total),
+                       ("X", X_loss), ("Y", Y_loss)):
+        gs = torch.autograd.grad(loss, gan.groups[name].params,
+                                 retain_graph=True)
+        want[name] = torch.cat([g.reshape(-1) for g in gs])
+
+    # --- production train_step (optimizer step leaves grads in place) ---
+    gan.train_step(x, y)
+    for name in ("G", "F", "X", "Y"):
+        got = gan.groups[name].flat_grad
+        assert torch.allclose(got, want[name], rtol=1e-5, atol=1e-7), name

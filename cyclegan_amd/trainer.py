@@ -83,6 +83,9 @@ class CycleGAN:
                 _shadow.register_arena(a)
                 self.arenas[name] = a
 
+        # lazily-captured hip-graph step (main.py at N=1)
+        self.graphed = None
+
     def _refresh_shadows(self):
         for a in self.arenas.values():
             a.refresh()
@@ -328,11 +331,19 @@ class GraphedStep:
     """
 
     def __init__(self, gan: CycleGAN, x: torch.Tensor, y: torch.Tensor,
-                 warmup: int = 2):
+                 warmup: int = 2, preserve_state: bool = False):
         assert gan.ctx.device.type == "cuda" and gan.ctx.world_size == 1
         self.gan = gan
         self.sx = gan._cast(x).clone()
         self.sy = gan._cast(y).clone()
+        # preserve_state: roll model/optimizer state back after the warmup
+        # steps so capturing mid-training perturbs nothing (main.py uses
+        # this to capture on the first batch of an epoch)
+        saved = None
+        if preserve_state:
+            saved = {n: (g.flat_param.clone(), gan.optimizers[n].m.clone(),
+                         gan.optimizers[n].v.clone(), gan.optimizers[n].t)
+                     for n, g in gan.groups.items()}
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
@@ -347,6 +358,15 @@ class GraphedStep:
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
             self.out = gan.train_step(self.sx, self.sy)
+            self._packed = torch.stack([self.out[k] for k in self.gan._TRAIN_KEYS])
+        if saved is not None:
+            with torch.no_grad():
+                for n, (p, m, v, t) in saved.items():
+                    gan.groups[n].flat_param.copy_(p)
+                    gan.optimizers[n].m.copy_(m)
+                    gan.optimizers[n].v.copy_(v)
+                    gan.optimizers[n].t = t
+                gan._refresh_shadows()
 
     def __call__(self, x=None, y=None) -> Dict[str, torch.Tensor]:
         if x is not None:
@@ -358,3 +378,10 @@ class GraphedStep:
         for g in self.gan.groups.values():
             g.bump_versions()
         return self.out
+
+    def call_cloned(self, x, y) -> Dict[str, torch.Tensor]:
+        """Replay and return loss scalars detached from the static graph
+        outputs (ONE clone kernel), safe to accumulate across steps."""
+        self(x, y)
+        p = self._packed.clone()
+        return {k: p[i] for i, k in enumerate(self.gan._TRAIN_KEYS)}

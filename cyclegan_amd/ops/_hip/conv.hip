@@ -1416,6 +1416,33 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   const int tr_row_a = jg >> 2;            // row-in-4 within the tr tile
   const int tr_cb_a = (jg & 3) * 8;        // 4-elem column sub-offset
 
+  // tr-read addresses are kk-invariant up to +kk*rowpitch: kk in {0,32}
+  // only touches address bits >= 5 of the row, so AXOR/DXOR (bits derived
+  // from row&7 and (row>>3)&1, with fg*8 keeping row&7 = tr_row_a and
+  // kk>>3 even) never change across the m-loop. Hoist the full per-lane
+  // byte offsets once; the inner loop is pure adds (was the 7.6 VALU/MFMA
+  // hot spot, profiles/pmc_bench286.md row 1).
+  int a_base[KF][2], d_base[NFW][2];
+  #pragma unroll
+  for (int kf = 0; kf < KF; ++kf) {
+    #pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      int row = fg * 8 + tr_row_a + 4 * h;
+      int cbl = (wk0 + kf * 16) * 2 + tr_cb_a;
+      a_base[kf][h] = row * 256 + (cbl ^ AXOR(row));
+    }
+  }
+  #pragma unroll
+  for (int nf = 0; nf < NFW; ++nf) {
+    #pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      int row = fg * 8 + tr_row_a + 4 * h;
+      int cbl = (wn0 + nf * 16) * 2 + tr_cb_a;
+      d_base[nf][h] = row * (WBN * 2) +
+          (cbl ^ (WBN == 128 ? AXOR(row) : WBN == 64 ? DXOR(row) : 0));
+    }
+  }
+
   stage(0, mstart);
   __syncthreads();
 
@@ -1433,25 +1460,15 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
       v8bf a[KF];
       #pragma unroll
       for (int kf = 0; kf < KF; ++kf) {
-        int cbl = (wk0 + kf * 16) * 2 + tr_cb_a;
-        int row = kk + fg * 8 + tr_row_a;
-        v4bfx lo = tr16_read(Ab + row * 256 + (cbl ^ AXOR(row)));
-        row += 4;
-        v4bfx hi = tr16_read(Ab + row * 256 + (cbl ^ AXOR(row)));
+        v4bfx lo = tr16_read(Ab + kk * 256 + a_base[kf][0]);
+        v4bfx hi = tr16_read(Ab + kk * 256 + a_base[kf][1]);
         a[kf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
       v8bf bfr[NFW];
       #pragma unroll
       for (int nf = 0; nf < NFW; ++nf) {
-        int cbl = (wn0 + nf * 16) * 2 + tr_cb_a;
-        int row = kk + fg * 8 + tr_row_a;
-        v4bfx lo = tr16_read(Db + row * (WBN * 2) +
-                             (cbl ^ (WBN == 128 ? AXOR(row)
-                                     : WBN == 64 ? DXOR(row) : 0)));
-        row += 4;
-        v4bfx hi = tr16_read(Db + row * (WBN * 2) +
-                             (cbl ^ (WBN == 128 ? AXOR(row)
-                                     : WBN == 64 ? DXOR(row) : 0)));
+        v4bfx lo = tr16_read(Db + kk * (WBN * 2) + d_base[nf][0]);
+        v4bfx hi = tr16_read(Db + kk * (WBN * 2) + d_base[nf][1]);
         bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
       #pragma unroll

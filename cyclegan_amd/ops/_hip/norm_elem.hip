@@ -203,9 +203,13 @@ __global__ __launch_bounds__(NT) void in_bwd_reduce_kernel(
     const short* __restrict__ yact, const float* __restrict__ mean,
     const float* __restrict__ rstd,
     float* __restrict__ p1, float* __restrict__ p2, int B, long HW, int C,
-    int S, int act, float slope) {
+    int S, int act, float slope, float* __restrict__ dgb) {
   int b = blockIdx.x / S;
   int sl = blockIdx.x % S;
+  // zero the dgamma/dbeta accumulator the NEXT kernel (in_bwd_dx) fills
+  // with atomics — saves the host-side at::zeros fill launch
+  if (blockIdx.x == 0)
+    for (int i = threadIdx.x; i < 2 * C; i += NT) dgb[i] = 0.f;
   long rows = (HW + S - 1) / S;
   long r0 = sl * rows, r1 = min(r0 + rows, HW);
   const int gpr = C / 8;
@@ -657,7 +661,7 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
   auto p1 = at::empty({S, B, C}, fopt);
   auto p2 = at::empty({S, B, C}, fopt);
   auto dx = at::empty_like(x);
-  auto dgb = at::zeros({2, C}, fopt);
+  auto dgb = at::empty({2, C}, fopt);  // zeroed by in_bwd_reduce_kernel
   auto dbeta = dgb[0];
   auto dgamma = dgb[1];
   const short* yp = yact.has_value()
@@ -669,7 +673,7 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
                      (const float*)rstd.const_data_ptr(),
                      (float*)p1.mutable_data_ptr(),
                      (float*)p2.mutable_data_ptr(), B, HW, C, S, (int)act,
-                     (float)slope);
+                     (float)slope, (float*)dgb.mutable_data_ptr());
   hipLaunchKernelGGL(in_bwd_dx_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)dy.const_data_ptr(),
                      (const short*)x.const_data_ptr(), yp,
@@ -691,7 +695,9 @@ at::Tensor channel_sum(at::Tensor x) {
   long rows = x.numel() / C;
   TORCH_CHECK(C % 8 == 0 && (C / 8) <= NT && NT % (C / 8) == 0,
               "channel_sum: unsupported C ", C);
-  auto out = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto out = at::empty({C}, x.options().dtype(at::kFloat));
+  CHECK_HIP(hipMemsetAsync(out.mutable_data_ptr(), 0, C * sizeof(float),
+                           at::cuda::getCurrentCUDAStream()));
   int S = (int)std::min<long>(512, std::max<long>(1, rows / 64));
   hipLaunchKernelGGL(channel_sum_kernel, dim3(S), dim3(NT), 0,
                      at::cuda::getCurrentCUDAStream(),
@@ -749,7 +755,9 @@ static at::Tensor persample_fwd_impl(c10::optional<at::Tensor> yt,
   TORCH_CHECK(yp.is_cuda() && yp.scalar_type() == at::kBFloat16 && yp.is_contiguous());
   int B = yp.size(0);
   long D = yp.numel() / B;
-  auto out = at::zeros({B}, yp.options().dtype(at::kFloat));
+  auto out = at::empty({B}, yp.options().dtype(at::kFloat));
+  CHECK_HIP(hipMemsetAsync(out.mutable_data_ptr(), 0, B * sizeof(float),
+                           at::cuda::getCurrentCUDAStream()));
   int gx = (int)std::min<long>(32, std::max<long>(1, D / (256 * 8)));
   dim3 grid(gx, B);
   hipLaunchKernelGGL(persample_loss_kernel, grid, dim3(NT), 0,

@@ -159,10 +159,12 @@ class CycleGAN:
         # F_total, whose grads stop at F's own weights), so cutting it is
         # gradient-identical and lets F run as ONE 3b-batched call without
         # autograd descending a numerically-zero path into G.
-        g_out = self.G(torch.cat([x, y]))     # -> fake_y, same_y
-        fake_y, same_y = g_out[:b], g_out[b:]
+        # torch.split, not slicing: split's backward is ONE cat of the
+        # branch grads; slice backward is a zero-fill + copy + add per
+        # branch (and leaves non-contiguous dy for the conv backwards)
+        fake_y, same_y = self.G(torch.cat([x, y])).split(b)
         f_out = self.F(torch.cat([y, x, fake_y.detach()]))
-        fake_x, same_x, cycle_x = f_out[:b], f_out[b:2 * b], f_out[2 * b:]
+        fake_x, same_x, cycle_x = f_out.split(b)
         cycle_y = self.G(fake_x.detach())
 
         # ONE discriminator pass per fake, shared by the adversarial and

@@ -1333,11 +1333,11 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
 
   auto full_a = [&](int j) {
     int ih = a_ih[j], iw = a_iw[j];
-    bool valid = true;
+    bool valid = a_kv[j];   // k-tail folds in permanently
     if (p.reflect) {
       ih = mirror_idx(ih, p.H); iw = mirror_idx(iw, p.W);
     } else {
-      valid = ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
+      valid = valid && ih >= 0 && ih < p.H && iw >= 0 && iw < p.W;
     }
     avalid[j] = valid;
     avo[j] = valid ? (unsigned)(((((long)a_b[j] * p.H + ih) * p.W + iw) *
@@ -1360,14 +1360,23 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   }
   #pragma unroll
   for (int j = 0; j < DPI; ++j)
-    dvo[j] = (unsigned)(((mstart + d_row[j]) * p.Cout + d_n[j]) * 2);
+    // n-overflow folds permanently: 0xFF000000 plus all increments stays
+    // far above num_records (tensor < 100 MB), so the load returns 0
+    dvo[j] = d_n[j] < p.Cout
+                 ? (unsigned)(((mstart + d_row[j]) * p.Cout + d_n[j]) * 2)
+                 : 0xFF000000u;
 
+  const bool ow_fast = (p.OW == WG_BM);  // K3/down shapes: ow invariant
   auto advance = [&]() {
     #pragma unroll
     for (int j = 0; j < API; ++j) {
       int ow0 = a_ow[j], oh0 = a_oh[j], b0 = a_b[j];
-      a_ow[j] += WG_BM;
-      while (a_ow[j] >= p.OW) { a_ow[j] -= p.OW; ++a_oh[j]; }
+      if (ow_fast) {
+        ++a_oh[j];
+      } else {
+        a_ow[j] += WG_BM;
+        while (a_ow[j] >= p.OW) { a_ow[j] -= p.OW; ++a_oh[j]; }
+      }
       while (a_oh[j] >= p.OH) { a_oh[j] -= p.OH; ++a_b[j]; }
       int dih = (a_oh[j] - oh0) * p.stride;
       int diw = (a_ow[j] - ow0) * p.stride;
@@ -1375,7 +1384,7 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
       a_iw[j] += diw;
       bool inb = (unsigned)a_ih[j] < (unsigned)p.H &&
                  (unsigned)a_iw[j] < (unsigned)p.W;
-      if (a_b[j] == b0 && inb && ainb[j]) {
+      if (a_b[j] == b0 && inb && ainb[j] && avalid[j]) {
         avo[j] += (unsigned)(dih * (int)WC2 + diw * (int)C2);
       } else {
         full_a(j);
@@ -1386,11 +1395,13 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     for (int j = 0; j < DPI; ++j) dvo[j] += (unsigned)(WG_BM * p.Cout * 2);
   };
 
-  auto stage = [&](int buf, long ms) {
+  // m-tail guard is UNIFORM (a_row < WG_BM): only the last chunk of a
+  // slice that crosses p.M needs per-row selects; the steady state stages
+  // straight from the folded avo/dvo registers.
+  auto stage = [&](int buf, long ms, bool tail) {
     #pragma unroll
     for (int j = 0; j < API; ++j) {
-      unsigned vo = (a_kv[j] && avalid[j] && ms + a_row[j] < p.M)
-                        ? avo[j] : 0xFF000000u;
+      unsigned vo = (!tail || ms + a_row[j] < p.M) ? avo[j] : 0xFF000000u;
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
           rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * API + j) * 512],
           16, vo, 0, 0, 0);
@@ -1398,8 +1409,7 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     if (dw_on) {
       #pragma unroll
       for (int j = 0; j < DPI; ++j) {
-        unsigned vo = (d_n[j] < p.Cout && ms + d_row[j] < p.M)
-                          ? dvo[j] : 0xFF000000u;
+        unsigned vo = (!tail || ms + d_row[j] < p.M) ? dvo[j] : 0xFF000000u;
         __builtin_amdgcn_raw_ptr_buffer_load_lds(
             rd, (__attribute__((address_space(3))) void*)&sm.D[buf][(w * DPI + j) * 512],
             16, vo, 0, 0, 0);
@@ -1443,14 +1453,14 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     }
   }
 
-  stage(0, mstart);
+  stage(0, mstart, mstart + WG_BM > p.M);
   __syncthreads();
 
   int cur = 0;
   for (long ms = mstart; ms < mend; ms += WG_BM, cur ^= 1) {
     if (ms + WG_BM < mend) {
       advance();
-      stage(cur ^ 1, ms + WG_BM);
+      stage(cur ^ 1, ms + WG_BM, ms + 2 * WG_BM > p.M);
     }
     const char* Ab = (const char*)sm.A[cur];
     const char* Db = (const char*)sm.D[cur];

@@ -1286,7 +1286,16 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   const long mstart = sl * p.mchunks_per_slice * WG_BM;
   long mend = mstart + p.mchunks_per_slice * WG_BM;
   if (mend > p.M) mend = p.M;
-  if (mstart >= mend) return;
+  if (mstart >= mend) {
+    // idle slice (host over-split): MUST still zero its output slab —
+    // wgrad_reduce sums every chunk, and at::empty memory is dirty
+    // (this was a real bug: reused allocator pages leaked garbage into
+    // dw at shapes where slices * mchunks_per_slice overshot M)
+    long chunk = (((long)sl * p.ktiles + kt) * p.ntiles + nt) *
+                 ((long)WBN * WG_BK);
+    for (int i = tid; i < WBN * WG_BK; i += NW * 64) p.ws[chunk + i] = 0.f;
+    return;
+  }
 
   auto rx = __builtin_amdgcn_make_buffer_rsrc(
       (void*)p.x, 0, (unsigned)((long)p.B * p.H * p.W * p.Cin * 2), 0x00020000);
@@ -1884,7 +1893,8 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
   int target = std::max<long>(1, tgt_blocks / ((long)p.ktiles * p.ntiles));
   int slices = (int)std::min<long>(mchunks, target);
   p.mchunks_per_slice = (mchunks + slices - 1) / slices;
-  p.slices = slices;
+  // re-derive so no slice is empty (ceil division can overshoot)
+  p.slices = (int)((mchunks + p.mchunks_per_slice - 1) / p.mchunks_per_slice);
   dim3 grid((long)p.ktiles * p.ntiles * p.slices);
   bool glds_ok = (p.Cin % 8) == 0 && (p.Cout % 8) == 0 &&
                  (long)p.B * p.H * p.W * p.Cin * 2 < (1L << 31) &&
@@ -1908,6 +1918,8 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
       int target = std::max<long>(1, tgt2 / ((long)p.ktiles * p.ntiles));
       p.slices = (int)std::min<long>(mchunks, target);
       p.mchunks_per_slice = (mchunks + p.slices - 1) / p.slices;
+      p.slices = (int)((mchunks + p.mchunks_per_slice - 1) /
+                       p.mchunks_per_slice);
       grid = dim3((long)p.ktiles * p.ntiles * p.slices);
     }
     auto ws = at::empty({(long)p.slices * p.ktiles * p.ntiles *

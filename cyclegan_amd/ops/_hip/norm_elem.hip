@@ -381,6 +381,299 @@ __global__ __launch_bounds__(NT) void in_bwd_dx_kernel(
   }
 }
 
+// ---- fused backward: partials + dx in ONE launch ----
+// The two-pass backward reads (dy, x, yact) from HBM twice. Here each
+// block computes its slice partials, publishes them with the CDNA4 G16
+// recipe (R1 fan-in): pair-granule agent-scope sc1 stores + vmcnt drain +
+// one arrival-counter bump; every block of the sample polls the ONE
+// counter word, takes ONE agent acquire, then reduces the partials with
+// plain pipelined loads and runs the dx pass re-reading its own rows
+// L2-hot. Grid B*S is clamped by an occupancy query on the host so every
+// block is co-resident; the poll is bounded (poison + give-up) so a
+// scheduling surprise cannot wedge the GPU.
+
+#define RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
+typedef __attribute__((address_space(1))) unsigned long long gu64;
+typedef __attribute__((address_space(1))) unsigned int gu32;
+
+__global__ __launch_bounds__(NT) void in_bwd_fused_kernel(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    const short* __restrict__ yact, const float* __restrict__ gamma,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    unsigned long long* __restrict__ gq,   // [S*B*C] {p2,p1} pair granules
+    unsigned int* __restrict__ arrive,     // [B] counters, zeroed per launch
+    short* __restrict__ dx, float* __restrict__ dbeta,
+    float* __restrict__ dgamma, int B, long HW, int C, int S, int act,
+    float slope) {
+  int b = blockIdx.x / S;
+  int sl = blockIdx.x % S;
+  long rows = (HW + S - 1) / S;
+  long r0 = sl * rows, r1 = min(r0 + rows, HW);
+  const int gpr = C / 8;
+  const int tid = threadIdx.x;
+  const long base = (long)b * HW * C;
+  const float inv_hw = 1.f / (float)HW;
+
+  // ---- phase 1: slice partials (same math as in_bwd_reduce_kernel) ----
+  int g = tid % gpr;
+  int rstep = NT / gpr;
+  int rof = tid / gpr;
+  float a1[8] = {}, a2[8] = {};
+  float mv[8], rv[8];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mv[j] = mean[(long)b * C + g * 8 + j];
+    rv[j] = rstd[(long)b * C + g * 8 + j];
+  }
+  #pragma unroll 4
+  for (long r = r0 + rof; r < r1; r += rstep) {
+    long off = base + r * C + g * 8;
+    v8s dv = *(const v8s*)(dy + off);
+    v8s xv = *(const v8s*)(x + off);
+    v8s yv = {};
+    if (act != ACT_NONE) yv = *(const v8s*)(yact + off);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float d = b2f(dv[j]);
+      if (act != ACT_NONE) d = actbw(d, b2f(yv[j]), act, slope);
+      float xh = (b2f(xv[j]) - mv[j]) * rv[j];
+      a1[j] += d; a2[j] += d * xh;
+    }
+  }
+  __shared__ float red[NT * 17];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[tid * 17 + j] = a1[j];
+    red[tid * 17 + 8 + j] = a2[j];
+  }
+  __syncthreads();
+  long gout = ((long)sl * B + b) * C;
+  for (int c = tid; c < C; c += NT) {
+    int g2 = c / 8, j = c % 8;
+    float t1 = 0, t2 = 0;
+    for (int k = 0; k < rstep; ++k) {
+      t1 += red[(g2 + k * gpr) * 17 + j];
+      t2 += red[(g2 + k * gpr) * 17 + 8 + j];
+    }
+    unsigned long long pack =
+        ((unsigned long long)__float_as_uint(t2) << 32) | __float_as_uint(t1);
+    __hip_atomic_store((gu64*)&gq[gout + c], pack, RLX_AGENT);  // sc1
+  }
+  // publish: drain the granule stores, then ONE lane bumps the counter
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (tid == 0)
+    __hip_atomic_fetch_add((gu32*)&arrive[b], 1u, RLX_AGENT);
+
+  // ---- phase 2: wait for all S slices of sample b, then reduce ----
+  __shared__ float sm1[MAXC], sm2[MAXC], smean[MAXC], srstd[MAXC];
+  __shared__ int failed;
+  if (tid == 0) {
+    failed = 0;
+    unsigned spins = 0;
+    while (__hip_atomic_load((gu32*)&arrive[b], RLX_AGENT) < (unsigned)S) {
+      __builtin_amdgcn_s_sleep(8);
+      if (++spins > (1u << 24)) { failed = 1; break; }  // ~bounded
+    }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");  // ONE acquire
+  }
+  __syncthreads();
+  if (failed) {  // co-residency violated: poison instead of hanging
+    for (long r = r0 + rof; r < r1; r += rstep) {
+      long off = base + r * C + g * 8;
+      v8s out;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) out[j] = f2b(__int_as_float(0x7FC00000));
+      *(v8s*)(dx + off) = out;
+    }
+    return;
+  }
+  bool sl0 = (sl == 0);
+  for (int c = tid; c < C; c += NT) {
+    float t1 = 0, t2 = 0;
+    int k = 0;
+    for (; k + 4 <= S; k += 4) {  // plain pipelined loads (post-acquire)
+      unsigned long long v0 = gq[((long)(k + 0) * B + b) * C + c];
+      unsigned long long v1 = gq[((long)(k + 1) * B + b) * C + c];
+      unsigned long long v2 = gq[((long)(k + 2) * B + b) * C + c];
+      unsigned long long v3 = gq[((long)(k + 3) * B + b) * C + c];
+      t1 += (__uint_as_float((unsigned)v0) + __uint_as_float((unsigned)v1)) +
+            (__uint_as_float((unsigned)v2) + __uint_as_float((unsigned)v3));
+      t2 += (__uint_as_float((unsigned)(v0 >> 32)) +
+             __uint_as_float((unsigned)(v1 >> 32))) +
+            (__uint_as_float((unsigned)(v2 >> 32)) +
+             __uint_as_float((unsigned)(v3 >> 32)));
+    }
+    for (; k < S; ++k) {
+      unsigned long long v = gq[((long)k * B + b) * C + c];
+      t1 += __uint_as_float((unsigned)v);
+      t2 += __uint_as_float((unsigned)(v >> 32));
+    }
+    sm1[c] = t1 * inv_hw;
+    sm2[c] = t2 * inv_hw;
+    smean[c] = mean[(long)b * C + c];
+    srstd[c] = rstd[(long)b * C + c];
+    if (sl0) {  // dgamma/dbeta fold: B adders per c (dgb memset on stream)
+      atomicAdd(&dbeta[c], t1);
+      atomicAdd(&dgamma[c], t2);
+    }
+  }
+  __syncthreads();
+
+  // ---- phase 3: dx over this block's rows (L2-hot re-read) ----
+  #pragma unroll 4
+  for (long r = r0 + rof; r < r1; r += rstep) {
+    long off = base + r * C + g * 8;
+    v8s dv = *(const v8s*)(dy + off);
+    v8s xv = *(const v8s*)(x + off);
+    v8s yv = {};
+    if (act != ACT_NONE) yv = *(const v8s*)(yact + off);
+    v8s out;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = g * 8 + j;
+      float rs = srstd[c];
+      float xh = (b2f(xv[j]) - smean[c]) * rs;
+      float d = b2f(dv[j]);
+      if (act != ACT_NONE) d = actbw(d, b2f(yv[j]), act, slope);
+      out[j] = f2b(gamma[c] * rs * (d - sm1[c] - xh * sm2[c]));
+    }
+    *(v8s*)(dx + off) = out;
+  }
+}
+
+// ---- fused forward: stats + normalize in ONE launch (G16 fan-in, same
+// scheme as in_bwd_fused_kernel: pair granules + arrival counter) ----
+__global__ __launch_bounds__(NT) void in_fwd_fused_kernel(
+    const short* __restrict__ x, const float* __restrict__ gamma,
+    const float* __restrict__ beta,
+    unsigned long long* __restrict__ gq,   // [S*B*C] {q,s} pair granules
+    unsigned int* __restrict__ arrive,     // [B] counters, zeroed per launch
+    float* __restrict__ mean, float* __restrict__ rstd,
+    const short* __restrict__ res, short* __restrict__ y, int B, long HW,
+    int C, int S, int act, float slope, float eps) {
+  int b = blockIdx.x / S;
+  int sl = blockIdx.x % S;
+  long rows = (HW + S - 1) / S;
+  long r0 = sl * rows, r1 = min(r0 + rows, HW);
+  const int gpr = C / 8;
+  const int tid = threadIdx.x;
+  const long base = (long)b * HW * C;
+  const float inv_hw = 1.f / (float)HW;
+
+  // ---- phase 1: slice partials (in_reduce body) ----
+  int g = tid % gpr;
+  int rstep = NT / gpr;
+  int rof = tid / gpr;
+  float s[8] = {}, q[8] = {};
+  #pragma unroll 4
+  for (long r = r0 + rof; r < r1; r += rstep) {
+    v8s v = *(const v8s*)(x + base + r * C + g * 8);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = b2f(v[j]);
+      s[j] += f; q[j] += f * f;
+    }
+  }
+  __shared__ float red[NT * 17];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[tid * 17 + j] = s[j];
+    red[tid * 17 + 8 + j] = q[j];
+  }
+  __syncthreads();
+  long gout = ((long)sl * B + b) * C;
+  for (int c = tid; c < C; c += NT) {
+    int g2 = c / 8, j = c % 8;
+    float ts = 0, tq = 0;
+    for (int k = 0; k < rstep; ++k) {
+      ts += red[(g2 + k * gpr) * 17 + j];
+      tq += red[(g2 + k * gpr) * 17 + 8 + j];
+    }
+    unsigned long long pack =
+        ((unsigned long long)__float_as_uint(tq) << 32) | __float_as_uint(ts);
+    __hip_atomic_store((gu64*)&gq[gout + c], pack, RLX_AGENT);  // sc1
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (tid == 0)
+    __hip_atomic_fetch_add((gu32*)&arrive[b], 1u, RLX_AGENT);
+
+  // ---- phase 2: wait + per-(b,c) stats ----
+  __shared__ float sm[MAXC], sr[MAXC];
+  __shared__ int failed;
+  if (tid == 0) {
+    failed = 0;
+    unsigned spins = 0;
+    while (__hip_atomic_load((gu32*)&arrive[b], RLX_AGENT) < (unsigned)S) {
+      __builtin_amdgcn_s_sleep(8);
+      if (++spins > (1u << 24)) { failed = 1; break; }
+    }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  }
+  __syncthreads();
+  if (failed) {
+    for (long r = r0 + rof; r < r1; r += rstep) {
+      long off = base + r * C + g * 8;
+      v8s out;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) out[j] = f2b(__int_as_float(0x7FC00000));
+      *(v8s*)(y + off) = out;
+    }
+    return;
+  }
+  for (int c = tid; c < C; c += NT) {
+    float sv = 0, qv = 0;
+    int k = 0;
+    for (; k + 4 <= S; k += 4) {
+      unsigned long long v0 = gq[((long)(k + 0) * B + b) * C + c];
+      unsigned long long v1 = gq[((long)(k + 1) * B + b) * C + c];
+      unsigned long long v2 = gq[((long)(k + 2) * B + b) * C + c];
+      unsigned long long v3 = gq[((long)(k + 3) * B + b) * C + c];
+      sv += (__uint_as_float((unsigned)v0) + __uint_as_float((unsigned)v1)) +
+            (__uint_as_float((unsigned)v2) + __uint_as_float((unsigned)v3));
+      qv += (__uint_as_float((unsigned)(v0 >> 32)) +
+             __uint_as_float((unsigned)(v1 >> 32))) +
+            (__uint_as_float((unsigned)(v2 >> 32)) +
+             __uint_as_float((unsigned)(v3 >> 32)));
+    }
+    for (; k < S; ++k) {
+      unsigned long long v = gq[((long)k * B + b) * C + c];
+      sv += __uint_as_float((unsigned)v);
+      qv += __uint_as_float((unsigned)(v >> 32));
+    }
+    float m = sv * inv_hw;
+    float var = qv * inv_hw - m * m;
+    if (var < 0.f) var = 0.f;
+    float rs = rsqrtf(var + eps);
+    sm[c] = m;
+    sr[c] = rs;
+    if (sl == 0) {
+      mean[(long)b * C + c] = m;
+      rstd[(long)b * C + c] = rs;
+    }
+  }
+  __syncthreads();
+
+  // ---- phase 3: normalize + affine + act + residual (L2-hot re-read) ----
+  #pragma unroll 4
+  for (long r = r0 + rof; r < r1; r += rstep) {
+    long off = base + r * C + g * 8;
+    v8s v = *(const v8s*)(x + off);
+    v8s rv = {};
+    if (res) rv = *(const v8s*)(res + off);
+    v8s out;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = g * 8 + j;
+      float val = (b2f(v[j]) - sm[c]) * sr[c] * gamma[c] + beta[c];
+      if (res) val += b2f(rv[j]);
+      out[j] = f2b(apply_act(val, act, slope));
+    }
+    *(v8s*)(y + off) = out;
+  }
+}
+
 // ---- channel sum: out[c] = sum over (b,h,w) of x[...,c] (bias grads) ----
 __global__ __launch_bounds__(NT) void channel_sum_kernel(
     const short* __restrict__ x, float* __restrict__ out, long rows, int C,
@@ -620,11 +913,45 @@ std::vector<at::Tensor> instnorm_fwd(at::Tensor x, at::Tensor gamma,
   auto stream = at::cuda::getCurrentCUDAStream();
   auto fopt = x.options().dtype(at::kFloat);
   int S = slices_for(HW, B);
-  auto psum = at::empty({S, B, C}, fopt);
-  auto psq = at::empty({S, B, C}, fopt);
   auto mean = at::empty({B, C}, fopt);
   auto rstd = at::empty({B, C}, fopt);
   auto y = at::empty_like(x);
+  const short* resf = residual.has_value()
+                          ? (const short*)residual->const_data_ptr() : nullptr;
+
+  static int fused_cap = []() {
+    if (const char* e = getenv("CYG_IN_FUSED"); e && e[0] == '0') return 0;
+    int bpc = 0, ncu = 0, dev = 0;
+    if (hipGetDevice(&dev) != hipSuccess) return 0;
+    if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
+            &bpc, (const void*)in_fwd_fused_kernel, NT, 0) != hipSuccess)
+      return 0;
+    if (hipDeviceGetAttribute(&ncu, hipDeviceAttributeMultiprocessorCount,
+                              dev) != hipSuccess)
+      return 0;
+    return bpc * ncu;
+  }();
+  if (fused_cap > 0 && B < fused_cap) {
+    int Sf = std::max(1, std::min(S, fused_cap / B));
+    auto gq = at::empty({(long)Sf * B * C}, x.options().dtype(at::kLong));
+    auto arr = at::empty({B}, x.options().dtype(at::kInt));
+    CHECK_HIP(hipMemsetAsync(arr.mutable_data_ptr(), 0, (size_t)B * 4,
+                             stream));
+    hipLaunchKernelGGL(in_fwd_fused_kernel, dim3(B * Sf), dim3(NT), 0,
+                       stream, (const short*)x.const_data_ptr(),
+                       (const float*)gamma.const_data_ptr(),
+                       (const float*)beta.const_data_ptr(),
+                       (unsigned long long*)gq.mutable_data_ptr(),
+                       (unsigned int*)arr.mutable_data_ptr(),
+                       (float*)mean.mutable_data_ptr(),
+                       (float*)rstd.mutable_data_ptr(), resf,
+                       (short*)y.mutable_data_ptr(), B, HW, C, Sf,
+                       (int)act, (float)slope, (float)eps);
+    return {y, mean, rstd};
+  }
+
+  auto psum = at::empty({S, B, C}, fopt);
+  auto psq = at::empty({S, B, C}, fopt);
   hipLaunchKernelGGL(in_reduce_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)x.const_data_ptr(),
                      (float*)psum.mutable_data_ptr(),
@@ -658,14 +985,55 @@ std::vector<at::Tensor> instnorm_bwd(at::Tensor dy, at::Tensor x,
   auto stream = at::cuda::getCurrentCUDAStream();
   auto fopt = x.options().dtype(at::kFloat);
   int S = slices_for(HW, B);
-  auto p1 = at::empty({S, B, C}, fopt);
-  auto p2 = at::empty({S, B, C}, fopt);
   auto dx = at::empty_like(x);
-  auto dgb = at::empty({2, C}, fopt);  // zeroed by in_bwd_reduce_kernel
+  auto dgb = at::empty({2, C}, fopt);
   auto dbeta = dgb[0];
   auto dgamma = dgb[1];
   const short* yp = yact.has_value()
                         ? (const short*)yact->const_data_ptr() : nullptr;
+
+  // fused single-launch path (G16 fan-in): saves the second HBM read of
+  // (dy, x, yact). Grid must be co-resident for the in-launch handoff —
+  // clamp S by the occupancy query; fall back to the 2-kernel path when
+  // the clamp would starve the grid.
+  static int fused_cap = []() {
+    if (const char* e = getenv("CYG_IN_FUSED"); e && e[0] == '0') return 0;
+    int bpc = 0, ncu = 0, dev = 0;
+    if (hipGetDevice(&dev) != hipSuccess) return 0;
+    if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
+            &bpc, (const void*)in_bwd_fused_kernel, NT, 0) != hipSuccess)
+      return 0;
+    if (hipDeviceGetAttribute(&ncu, hipDeviceAttributeMultiprocessorCount,
+                              dev) != hipSuccess)
+      return 0;
+    return bpc * ncu;
+  }();
+  if (fused_cap > 0 && B < fused_cap) {
+    int Sf = std::max(1, std::min(S, fused_cap / B));
+    auto gq = at::empty({(long)Sf * B * C},
+                        x.options().dtype(at::kLong));
+    auto arr = at::empty({B}, x.options().dtype(at::kInt));
+    CHECK_HIP(hipMemsetAsync(arr.mutable_data_ptr(), 0, (size_t)B * 4,
+                             stream));
+    CHECK_HIP(hipMemsetAsync(dgb.mutable_data_ptr(), 0,
+                             (size_t)2 * C * 4, stream));
+    hipLaunchKernelGGL(in_bwd_fused_kernel, dim3(B * Sf), dim3(NT), 0,
+                       stream, (const short*)dy.const_data_ptr(),
+                       (const short*)x.const_data_ptr(), yp,
+                       (const float*)gamma.const_data_ptr(),
+                       (const float*)mean.const_data_ptr(),
+                       (const float*)rstd.const_data_ptr(),
+                       (unsigned long long*)gq.mutable_data_ptr(),
+                       (unsigned int*)arr.mutable_data_ptr(),
+                       (short*)dx.mutable_data_ptr(),
+                       (float*)dbeta.mutable_data_ptr(),
+                       (float*)dgamma.mutable_data_ptr(), B, HW, C, Sf,
+                       (int)act, (float)slope);
+    return {dx, dgamma, dbeta};
+  }
+
+  auto p1 = at::empty({S, B, C}, fopt);
+  auto p2 = at::empty({S, B, C}, fopt);
   hipLaunchKernelGGL(in_bwd_reduce_kernel, dim3(B * S), dim3(NT), 0, stream,
                      (const short*)dy.const_data_ptr(),
                      (const short*)x.const_data_ptr(), yp,

@@ -67,6 +67,44 @@ def _idx_bias(base: int, n: int) -> torch.Tensor:
     return idx.to(torch.int32)
 
 
+def head_packable(shape, stride, pad_mode, padding) -> bool:
+    """Generator-head geometry: 7x7 stride-1 reflect(3) conv with tiny
+    Cout and Cin divisible by 8 — runs as the 8-pixel-packed conv (see
+    ops.conv._ConvHeadPackedFn)."""
+    O, KH, KW, I = shape
+    return (O <= 8 and KH == 7 and KW == 7 and stride == 1 and I % 8 == 0
+            and I >= 8 and pad_mode == "reflect"
+            and tuple(padding) == (3, 3, 3, 3))
+
+
+def _idx_packp(base: int, shape) -> torch.Tensor:
+    """Packed head weight [8*8, KH, 2, 8*I]: n = d*8 + co packs 8 adjacent
+    output pixels into channels; k covers a 2-block (16-col) window; slot
+    (d, co, ty, blk, pos, ci) maps to w[co][ty][blk*8+pos-d][ci] when the
+    tap is in range, else zero (idx -1)."""
+    O, KH, KW, I = shape
+    assert KW == 7
+    d = torch.arange(8).view(-1, 1, 1, 1, 1, 1)
+    co = torch.arange(8).view(1, -1, 1, 1, 1, 1)
+    ty = torch.arange(KH).view(1, 1, -1, 1, 1, 1)
+    blk = torch.arange(2).view(1, 1, 1, -1, 1, 1)
+    pos = torch.arange(8).view(1, 1, 1, 1, -1, 1)
+    ci = torch.arange(I).view(1, 1, 1, 1, 1, -1)
+    tx = blk * 8 + pos - d
+    idx = base + ((co * KH + ty) * KW + tx) * I + ci
+    ok = (tx >= 0) & (tx < KW) & (co < O)
+    idx = torch.where(ok, idx, torch.tensor(-1))
+    return idx.reshape(64, KH, 2, 8 * I).to(torch.int32)
+
+
+def _idx_packb(base: int, n: int) -> torch.Tensor:
+    """Packed head bias [64]: b[d*8+co] = bias[co] (co < n), else 0."""
+    co = torch.arange(8).repeat(8)
+    idx = base + co
+    idx[co >= n] = -1
+    return idx.to(torch.int32)
+
+
 class ShadowArena:
     """Builds and refreshes the arena for one FlatParamGroup + its module."""
 
@@ -86,6 +124,12 @@ class ShadowArena:
             else:
                 continue
             w = m.weight
+            forms = list(forms)
+            packed = (_conv_like(m) and head_packable(
+                tuple(w.shape), m.stride, m.pad_mode, m.padding
+                if not isinstance(m.padding, str) else (0,) * 4))
+            if packed:
+                forms.append(("packp", _idx_packp))
             for name, fn in forms:
                 idx = fn(off[w], tuple(w.shape))
                 pieces.append(idx.reshape(-1))
@@ -96,6 +140,11 @@ class ShadowArena:
                 pieces.append(idx.reshape(-1))
                 plan.append((m.bias, "bias_p", tuple(idx.shape), total))
                 total += idx.numel()
+                if packed:
+                    idx = _idx_packb(off[m.bias], m.bias.numel())
+                    pieces.append(idx.reshape(-1))
+                    plan.append((m.bias, "packb", tuple(idx.shape), total))
+                    total += idx.numel()
 
         pad = (-total) % 8
         if pad:

@@ -34,7 +34,8 @@ import torch.nn.functional as F
 
 from . import backend
 from .shadow import (compute_weight, compute_weight_t, compute_weight_p,
-                     compute_weight_tp, compute_bias_p, fp8_weight_shadow)
+                     compute_weight_tp, compute_bias_p, compute_weight_packp,
+                     compute_bias_packb, fp8_weight_shadow)
 
 ACT_NONE, ACT_RELU, ACT_LRELU, ACT_TANH = 0, 1, 2, 3
 
@@ -156,6 +157,83 @@ class _ConvFn(torch.autograd.Function):
         return dx, dw, db, None, None, None, None, None
 
 
+def _head_packable(x, w, stride, pads, reflect) -> bool:
+    """Generator-head geometry (7x7 s1 reflect-3, Cout <= 8, Cin % 8 == 0):
+    runs as the 8-pixel-packed conv — 8 adjacent output pixels fold into
+    the MFMA N dimension (Cout' = 64) so the matrix cores run dense
+    instead of 8/16-wide, and the 49x input re-read of the skinny path
+    collapses into a 7x2-block window over the channel-folded image."""
+    import os
+    if os.environ.get("CYG_NO_PACKHEAD") == "1":
+        return False
+    return (reflect and stride == 1 and pads == (3, 3, 3, 3)
+            and w.shape[0] <= 8 and w.shape[1] == 7 and w.shape[2] == 7
+            and w.shape[3] % 8 == 0 and w.shape[3] >= 8
+            and x.shape[2] % 8 == 0 and x.shape[1] >= 7 and x.shape[2] >= 16)
+
+
+def _fold_packed_dw(dwp: torch.Tensor, wshape) -> torch.Tensor:
+    """[64, KH, 2, 8*I] packed wgrad -> OHWI [O, KH, KW, I]."""
+    O, KH, KW, I = wshape
+    d6 = dwp.view(8, 8, KH, 2, 8, I)                    # [d,co,ty,blk,pos,ci]
+    d7 = d6.permute(0, 3, 4, 1, 2, 5).reshape(8, 16, 8, KH, I)  # [d,a,co,ty,ci]
+    dev = dwp.device
+    idx = (torch.arange(KW, device=dev).view(1, -1)
+           + torch.arange(8, device=dev).view(-1, 1))   # a = tx + d
+    idx = idx.view(8, KW, 1, 1, 1).expand(8, KW, 8, KH, I)
+    dw = d7.gather(1, idx).sum(0)                       # [tx, co, ty, ci]
+    return dw.permute(1, 2, 0, 3)[:O].contiguous()      # [co,ty,tx,ci]
+
+
+class _ConvHeadPackedFn(torch.autograd.Function):
+    """8-pixel-packed head conv (see _head_packable). Forward and wgrad
+    run the packed dense-N GEMM; dgrad keeps the unpacked formulation
+    (its packed form would do 2.4x the FLOPs on zero weights)."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias, act, slope):
+        ext = backend.ext()
+        B, H, W, Cin = x.shape
+        cout = w.shape[0]
+        wpk = compute_weight_packp(w, x)
+        bpk = compute_bias_packb(bias, x) if bias is not None else None
+        xp = ext.reflect_pad_fwd(x.contiguous(), 3, 3, 3, 5)
+        xf = xp.view(B, H + 6, (W + 8) // 8, 8 * Cin)
+        yp = ext.conv2d_fwd(xf, wpk, bpk, 1, 0, 0, 0, 0, False, act, slope)
+        y = yp.view(B, H, W, 8)
+        if cout < 8:
+            y = y[..., :cout].contiguous()
+        ctx.save_for_backward(xp, w, y)
+        ctx.conf = (act, slope, bias is not None, cout, Cin, H, W)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        xp, w, y = ctx.saved_tensors
+        act, slope, has_bias, cout, Cin, H, W = ctx.conf
+        ext = backend.ext()
+        dy = dy.contiguous()
+        if act != ACT_NONE:
+            dy = ext.act_bwd(dy, y, act, slope)
+        dyp = _pad_channels(dy).contiguous()
+        db = None
+        if has_bias and ctx.needs_input_grad[2]:
+            db = ext.channel_sum(dyp)[: dy.shape[-1]]
+        dx = dw = None
+        B = dyp.shape[0]
+        if ctx.needs_input_grad[0]:
+            wt = compute_weight_tp(w, xp)
+            dx = ext.conv2d_dgrad(dyp, wt, H, W, 1, 3, 3, 3, 3, True)
+        if ctx.needs_input_grad[1]:
+            xf = xp.view(B, H + 6, (W + 8) // 8, 8 * Cin)
+            dy_pk = dyp.view(B, H, W // 8, 64)
+            dwp = ext.conv2d_wgrad(xf, dy_pk, 7, 2, 1, 0, 0, False)
+            dw = _fold_packed_dw(dwp, tuple(w.shape))
+            if dw.dtype != w.dtype:
+                dw = dw.to(w.dtype)
+        return dx, dw, db, None, None
+
+
 _FP8_MODE = False
 
 
@@ -256,6 +334,8 @@ def conv2d(
                 and max(w.shape[3], 8) % 16 == 0):
             return _ConvFp8Fn.apply(x, w, bias, stride, pads,
                                     pad_mode == "reflect", a, slope)
+        if _head_packable(x, w, stride, pads, pad_mode == "reflect"):
+            return _ConvHeadPackedFn.apply(x, w, bias, a, slope)
         return _ConvFn.apply(x, w, bias, stride, pads, pad_mode == "reflect", a, slope)
     wc = w if w.dtype == x.dtype else w.to(x.dtype)
     bc = bias if (bias is None or bias.dtype == x.dtype) else bias.to(x.dtype)

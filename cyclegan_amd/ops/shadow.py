@@ -145,6 +145,52 @@ def compute_weight_t(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
     return wt
 
 
+def _pack_head_weight_torch(w: torch.Tensor) -> torch.Tensor:
+    """Packed head form [64, KH, 2, 8*I] (see arena._idx_packp) built in
+    torch — fallback for masters without an arena (tests)."""
+    O, KH, KW, I = w.shape
+    out = w.new_zeros(8, 8, KH, 2, 8, I)
+    for d in range(8):
+        for blk in range(2):
+            for pos in range(8):
+                tx = blk * 8 + pos - d
+                if 0 <= tx < KW:
+                    out[d, :O, :, blk, pos, :] = w[:, :, tx, :]
+    return out.reshape(64, KH, 2, 8 * I)
+
+
+_cache_packp: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
+_cache_packb: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
+
+
+def compute_weight_packp(w: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
+    v = _arena(w, "packp", like)
+    if v is not None:
+        return v
+    ent = _cache_packp.get(w)
+    ver = w._version
+    if ent is not None and ent[0] == ver:
+        return ent[1]
+    wp = _pack_head_weight_torch(w.detach()).to(like.dtype).contiguous()
+    _cache_packp[w] = (ver, wp)
+    return wp
+
+
+def compute_bias_packb(b: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
+    v = _arena(b, "packb", like)
+    if v is not None:
+        return v
+    ent = _cache_packb.get(b)
+    ver = b._version
+    if ent is not None and ent[0] == ver:
+        return ent[1]
+    n = b.numel()
+    bp = torch.nn.functional.pad(b.detach(), (0, 8 - n)) if n < 8 else b.detach()
+    bp = bp.repeat(8).to(like.dtype).contiguous()
+    _cache_packb[b] = (ver, bp)
+    return bp
+
+
 def fp8_weight_shadow(w: torch.Tensor):
     """(wq uint8 e4m3 padded OHWI, sw fp32 scale tensor), cached per master
     version. scale = 448/amax (OCP e4m3 max normal)."""

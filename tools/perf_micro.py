@@ -89,6 +89,26 @@ cases += [
     ("head wgrad  ", fl5, lambda: e.conv2d_wgrad(x5, dy5, 7, 7, 1, 3, 3, True)),
 ]
 
+# packed head (8-pixel N-packing): fwd via reflect_pad + folded conv,
+# wgrad on the folded image (ops.conv._ConvHeadPackedFn path)
+from cyclegan_amd.ops.shadow import _pack_head_weight_torch
+w5pk = _pack_head_weight_torch(w5.float()).to(torch.bfloat16).contiguous()
+xp5 = e.reflect_pad_fwd(x5, 3, 3, 3, 5)
+xf5 = xp5.view(B, 262, 33, 512)
+dypk5 = dy5.view(B, 256, 32, 64)
+
+
+def head_packed_fwd():
+    xp = e.reflect_pad_fwd(x5, 3, 3, 3, 5)
+    xf = xp.view(B, 262, 33, 512)
+    return e.conv2d_fwd(xf, w5pk, None, 1, 0, 0, 0, 0, False, 0, 0.2)
+
+
+cases += [
+    ("headP fwd   ", fl5, head_packed_fwd),
+    ("headP wgrad ", fl5, lambda: e.conv2d_wgrad(xf5, dypk5, 7, 2, 1, 0, 0, False)),
+]
+
 # InstanceNorm 64^2 x 256
 g = torch.rand(256, device=DEV)
 bta = torch.rand(256, device=DEV)

@@ -16,6 +16,7 @@ at::Tensor conv2d_wgrad(at::Tensor, at::Tensor, int64_t, int64_t, int64_t,
                         int64_t, int64_t, bool);
 at::Tensor mfma_probe(at::Tensor, at::Tensor);
 at::Tensor mx_probe(at::Tensor, at::Tensor, int64_t, int64_t);
+at::Tensor mx_probe16(at::Tensor, at::Tensor, int64_t, int64_t);
 at::Tensor glds_probe(at::Tensor, at::Tensor);
 at::Tensor tr_probe(int64_t);
 at::Tensor quant_fp8(at::Tensor, at::Tensor);
@@ -52,6 +53,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_wgrad", &cyg::conv2d_wgrad);
   m.def("mfma_probe", &cyg::mfma_probe);
   m.def("mx_probe", &cyg::mx_probe);
+  m.def("mx_probe16", &cyg::mx_probe16);
   m.def("glds_probe", &cyg::glds_probe);
   m.def("tr_probe", &cyg::tr_probe);
   m.def("quant_fp8", &cyg::quant_fp8);

@@ -2100,6 +2100,33 @@ __global__ void mx_probe_kernel(const int* a, const int* b, float* d,
   for (int r = 0; r < 16; ++r) d[lane * 16 + r] = acc[r];
 }
 
+// 16x16x128 scaled-fp8 probe: same raw per-lane operand dump for the
+// D[16,16] = A[16,128] @ B[128,16] fragment (v4f acc).
+__global__ void mx_probe16_kernel(const int* a, const int* b, float* d,
+                                  int sa, int sb) {
+  int lane = threadIdx.x & 63;
+  v8i_ av = *(const v8i_*)(a + lane * 8);
+  v8i_ bv = *(const v8i_*)(b + lane * 8);
+  v4f acc = {};
+  acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+      av, bv, acc, 0, 0, 0, sa, 0, sb);
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) d[lane * 4 + r] = acc[r];
+}
+
+at::Tensor mx_probe16(at::Tensor a, at::Tensor b, int64_t sa, int64_t sb) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kInt &&
+              a.sizes() == at::IntArrayRef({64, 8}) &&
+              b.sizes() == at::IntArrayRef({64, 8}));
+  auto d = at::empty({64, 4}, a.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(mx_probe16_kernel, dim3(1), dim3(64), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     (const int*)a.contiguous().const_data_ptr(),
+                     (const int*)b.contiguous().const_data_ptr(),
+                     (float*)d.mutable_data_ptr(), (int)sa, (int)sb);
+  return d;
+}
+
 at::Tensor mx_probe(at::Tensor a, at::Tensor b, int64_t sa, int64_t sb) {
   TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kInt &&
               a.sizes() == at::IntArrayRef({64, 8}) &&

@@ -720,6 +720,8 @@ __global__ __launch_bounds__(NW * 64) void conv_halo_kernel(ConvParams p) {
 #include <hip/hip_fp8.h>
 
 constexpr int BKF = 128;  // fp8 K-step (elements == bytes)
+typedef int v4i __attribute__((ext_vector_type(4)));
+typedef int v8i __attribute__((ext_vector_type(8)));
 
 struct Fp8Smem {
   unsigned char A[2][BM * BKF];
@@ -904,23 +906,37 @@ __global__ __launch_bounds__(NTHREADS) void conv_fp8_kernel(ConvParams p) {
     }
     const char* Ab = (const char*)sm.A[kt & 1];
     const char* Bb = (const char*)sm.Bt[kt & 1];
-    #pragma unroll
-    for (int kk = 0; kk < BKF; kk += 32) {
-      const int kbyte = kk + fg * 8;
-      long a0 = *(const long*)(Ab + ((wm0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
-      long a1 = *(const long*)(Ab + ((wm0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
-      long a2 = *(const long*)(Ab + ((wm0 + 2 * 16 + fr) << 7) + (kbyte ^ swz));
-      long a3 = *(const long*)(Ab + ((wm0 + 3 * 16 + fr) << 7) + (kbyte ^ swz));
-      long b0 = *(const long*)(Bb + ((wn0 + 0 * 16 + fr) << 7) + (kbyte ^ swz));
-      long b1 = *(const long*)(Bb + ((wn0 + 1 * 16 + fr) << 7) + (kbyte ^ swz));
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b0, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b1, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a1, b0, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a1, b1, acc[1][1], 0, 0, 0);
-      acc[2][0] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a2, b0, acc[2][0], 0, 0, 0);
-      acc[2][1] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a2, b1, acc[2][1], 0, 0, 0);
-      acc[3][0] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a3, b0, acc[3][0], 0, 0, 0);
-      acc[3][1] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a3, b1, acc[3][1], 0, 0, 0);
+    // ONE mfma_scale_f32_16x16x128_f8f6f4 consumes the whole 128-B K-row
+    // (4x fewer matrix instructions than the 16x16x32 form). Fragment
+    // (mapped empirically, tools/mx_map.py map16): lane l supplies
+    // A[row = l%16][k = (l//16)*32 + b] for b = 0..31 — i.e. rows fr,
+    // byte window fg*32..fg*32+31, read as two XOR-swizzled b128s.
+    // Per-tensor scales are folded in the fp32 epilogue; sa = sb = 127
+    // (E8M0 2^0) keeps the MFMA unscaled.
+    {
+      const int c0 = (fg * 32) ^ swz;
+      const int c1 = (fg * 32 + 16) ^ swz;
+      v8i fa[4], fb[2];
+      #pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        const char* r = Ab + ((wm0 + mf * 16 + fr) << 7);
+        v4i lo = *(const v4i*)(r + c0);
+        v4i hi = *(const v4i*)(r + c1);
+        fa[mf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+      #pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        const char* r = Bb + ((wn0 + nf * 16 + fr) << 7);
+        v4i lo = *(const v4i*)(r + c0);
+        v4i hi = *(const v4i*)(r + c1);
+        fb[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+      #pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+        #pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              fa[mf], fb[nf], acc[mf][nf], 0, 0, 0, 127, 0, 127);
     }
     __syncthreads();
   }

@@ -1353,7 +1353,7 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   const unsigned WC2 = (unsigned)p.W * C2;
   int a_ow[API], a_oh[API], a_b[API], a_ih[API], a_iw[API];
   unsigned avo[API];
-  bool ainb[API], avalid[API];
+  bool ainb[API], avalid[API], a_iw_ok[API];
   unsigned dvo[DPI];
 
   auto full_a = [&](int j) {
@@ -1379,8 +1379,8 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     a_b[j] = t / p.OH;
     a_ih[j] = a_oh[j] * p.stride - p.pt + a_dkh[j];
     a_iw[j] = a_ow[j] * p.stride - p.pl + a_dkw[j];
-    ainb[j] = (unsigned)a_ih[j] < (unsigned)p.H &&
-              (unsigned)a_iw[j] < (unsigned)p.W;
+    a_iw_ok[j] = (unsigned)a_iw[j] < (unsigned)p.W;
+    ainb[j] = (unsigned)a_ih[j] < (unsigned)p.H && a_iw_ok[j];
     full_a(j);
   }
   #pragma unroll
@@ -1392,29 +1392,54 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
                  : 0xFF000000u;
 
   const bool ow_fast = (p.OW == WG_BM);  // K3/down shapes: ow invariant
+  const unsigned sWC2 = (unsigned)(p.stride * (int)WC2);
   auto advance = [&]() {
-    #pragma unroll
-    for (int j = 0; j < API; ++j) {
-      int ow0 = a_ow[j], oh0 = a_oh[j], b0 = a_b[j];
-      if (ow_fast) {
-        ++a_oh[j];
-      } else {
+    if (ow_fast) {
+      // ow (and iw) never change: interior step = oh+1, ih+stride, ONE
+      // voffset add. Batch wraps / ih border transitions take the rare
+      // exec-masked path (1/OH of steps) — this was the 20-mult-per-iter
+      // hot spot the compiler was predicating on every iteration.
+      #pragma unroll
+      for (int j = 0; j < API; ++j) {
+        bool rare = (++a_oh[j] >= p.OH);
+        if (!rare) {
+          int ih = a_ih[j] + p.stride;
+          a_ih[j] = ih;
+          bool inb = (unsigned)ih < (unsigned)p.H;
+          if (inb && ainb[j] && avalid[j]) {
+            avo[j] += sWC2;
+          } else {
+            rare = true;
+          }
+          ainb[j] = inb && a_iw_ok[j];
+        }
+        if (rare) {
+          while (a_oh[j] >= p.OH) { a_oh[j] -= p.OH; ++a_b[j]; }
+          a_ih[j] = a_oh[j] * p.stride - p.pt + a_dkh[j];
+          full_a(j);
+          ainb[j] = ((unsigned)a_ih[j] < (unsigned)p.H) && a_iw_ok[j];
+        }
+      }
+    } else {
+      #pragma unroll
+      for (int j = 0; j < API; ++j) {
+        int ow0 = a_ow[j], oh0 = a_oh[j], b0 = a_b[j];
         a_ow[j] += WG_BM;
         while (a_ow[j] >= p.OW) { a_ow[j] -= p.OW; ++a_oh[j]; }
+        while (a_oh[j] >= p.OH) { a_oh[j] -= p.OH; ++a_b[j]; }
+        int dih = (a_oh[j] - oh0) * p.stride;
+        int diw = (a_ow[j] - ow0) * p.stride;
+        a_ih[j] += dih;
+        a_iw[j] += diw;
+        bool inb = (unsigned)a_ih[j] < (unsigned)p.H &&
+                   (unsigned)a_iw[j] < (unsigned)p.W;
+        if (a_b[j] == b0 && inb && ainb[j] && avalid[j]) {
+          avo[j] += (unsigned)(dih * (int)WC2 + diw * (int)C2);
+        } else {
+          full_a(j);
+        }
+        ainb[j] = inb;
       }
-      while (a_oh[j] >= p.OH) { a_oh[j] -= p.OH; ++a_b[j]; }
-      int dih = (a_oh[j] - oh0) * p.stride;
-      int diw = (a_ow[j] - ow0) * p.stride;
-      a_ih[j] += dih;
-      a_iw[j] += diw;
-      bool inb = (unsigned)a_ih[j] < (unsigned)p.H &&
-                 (unsigned)a_iw[j] < (unsigned)p.W;
-      if (a_b[j] == b0 && inb && ainb[j] && avalid[j]) {
-        avo[j] += (unsigned)(dih * (int)WC2 + diw * (int)C2);
-      } else {
-        full_a(j);
-      }
-      ainb[j] = inb;
     }
     #pragma unroll
     for (int j = 0; j < DPI; ++j) dvo[j] += (unsigned)(WG_BM * p.Cout * 2);

@@ -20,7 +20,10 @@ at::Tensor mx_probe16(at::Tensor, at::Tensor, int64_t, int64_t);
 at::Tensor glds_probe(at::Tensor, at::Tensor);
 at::Tensor tr_probe(int64_t);
 at::Tensor quant_fp8(at::Tensor, at::Tensor);
-at::Tensor conv2d_fp8_fwd(at::Tensor, at::Tensor, at::Tensor, c10::optional<at::Tensor>,
+at::Tensor quant_fp8_d(at::Tensor, at::Tensor, at::Tensor);
+void amax_roll(at::Tensor, int64_t);
+at::Tensor conv2d_fp8_fwd(at::Tensor, at::Tensor, at::Tensor,
+                          c10::optional<at::Tensor>, c10::optional<at::Tensor>,
                           int64_t, int64_t, int64_t, int64_t, int64_t, bool, int64_t, double);
 std::vector<at::Tensor> instnorm_fwd(at::Tensor, at::Tensor, at::Tensor,
                                      double, int64_t, double,
@@ -57,6 +60,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("glds_probe", &cyg::glds_probe);
   m.def("tr_probe", &cyg::tr_probe);
   m.def("quant_fp8", &cyg::quant_fp8);
+  m.def("quant_fp8_d", &cyg::quant_fp8_d);
+  m.def("amax_roll", &cyg::amax_roll);
   m.def("conv2d_fp8_fwd", &cyg::conv2d_fp8_fwd);
   m.def("instnorm_fwd", &cyg::instnorm_fwd);
   m.def("instnorm_bwd", &cyg::instnorm_bwd);

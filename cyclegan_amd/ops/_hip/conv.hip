@@ -44,7 +44,8 @@ struct ConvParams {
   int stride, pt, pl;
   int reflect;                   // conv_fwd only
   int act;  float slope;
-  const float* dq;               // fp8 path: [0] = 1/(sx*sw)
+  const float* dq;               // fp8: amax_prev (with dqb) or 1/(sx*sw)
+  const float* dqb;              // fp8 delayed-scaling: sw scalar
   long M, KTOT;
   int mtiles, ntiles;
 };
@@ -754,6 +755,61 @@ __global__ void quant_fp8_kernel(const short* __restrict__ x,
   }
 }
 
+// ---- delayed-scaling quant ----
+// sx comes from the PREVIOUS step's amax (persistent per-layer slot,
+// transformer-engine style) so no separate full-tensor reduction runs
+// before quantization; this call's max|x| block-reduces and atomically
+// maxes into the CURRENT slot (nonneg float bits order == value order),
+// rolled once per step by amax_roll.
+DEV float delayed_sx(float amax_prev) {
+  return fminf(448.f / fmaxf(amax_prev, 1e-12f), 65504.f);
+}
+
+__global__ __launch_bounds__(256) void quant_fp8_d_kernel(
+    const short* __restrict__ x, const float* __restrict__ amax_prev,
+    float* __restrict__ amax_cur, unsigned char* __restrict__ y, long n) {
+  long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8;
+  const float s = delayed_sx(amax_prev[0]);
+  float mx = 0.f;
+  if (i + 8 <= n) {
+    v8s v = *(const v8s*)(x + i);
+    unsigned char out[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = b2f(v[j]);
+      mx = fmaxf(mx, fabsf(f));
+      __hip_fp8_e4m3 q(f * s);
+      out[j] = q.__x;
+    }
+    *(uint2*)(y + i) = *(uint2*)out;
+  } else if (i < n) {
+    for (long k = i; k < n; ++k) {
+      float f = b2f(x[k]);
+      mx = fmaxf(mx, fabsf(f));
+      __hip_fp8_e4m3 q(f * s);
+      y[k] = q.__x;
+    }
+  }
+  __shared__ float red[256];
+  red[threadIdx.x] = mx;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (threadIdx.x < off)
+      red[threadIdx.x] = fmaxf(red[threadIdx.x], red[threadIdx.x + off]);
+    __syncthreads();
+  }
+  if (threadIdx.x == 0)
+    atomicMax((unsigned*)amax_cur, __float_as_uint(red[0]));
+}
+
+__global__ void amax_roll_kernel(float* __restrict__ arena, int n, int cap) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    arena[i] = arena[cap + i];
+    arena[cap + i] = 0.f;
+  }
+}
+
 template <int STRIDE>
 __global__ __launch_bounds__(NTHREADS) void conv_fp8_kernel(ConvParams p) {
   const int stride = STRIDE ? STRIDE : p.stride;
@@ -941,7 +997,9 @@ __global__ __launch_bounds__(NTHREADS) void conv_fp8_kernel(ConvParams p) {
     __syncthreads();
   }
 
-  const float dqs = p.dq ? p.dq[0] : 1.f;
+  const float dqs = p.dq
+      ? (p.dqb ? 1.f / (delayed_sx(p.dq[0]) * p.dqb[0]) : p.dq[0])
+      : 1.f;
   #pragma unroll
   for (int nf = 0; nf < 2; ++nf) {
     int n = n0 + wn0 + nf * 16 + fr;
@@ -2017,7 +2075,37 @@ at::Tensor quant_fp8(at::Tensor x, at::Tensor scale) {
   return y;
 }
 
+at::Tensor quant_fp8_d(at::Tensor x, at::Tensor amax_prev,
+                       at::Tensor amax_cur) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
+              x.is_contiguous());
+  TORCH_CHECK(amax_prev.scalar_type() == at::kFloat &&
+              amax_cur.scalar_type() == at::kFloat);
+  auto y = at::empty_like(x, x.options().dtype(at::kByte));
+  long n = x.numel();
+  long blocks = (n / 8 + 255) / 256 + 1;
+  hipLaunchKernelGGL(quant_fp8_d_kernel, dim3((unsigned)blocks), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     (const short*)x.const_data_ptr(),
+                     (const float*)amax_prev.const_data_ptr(),
+                     (float*)amax_cur.mutable_data_ptr(),
+                     (unsigned char*)y.mutable_data_ptr(), n);
+  return y;
+}
+
+void amax_roll(at::Tensor arena, int64_t n) {
+  // arena [2, cap] fp32: prev <- cur; cur <- 0 for the first n slots
+  TORCH_CHECK(arena.is_cuda() && arena.scalar_type() == at::kFloat &&
+              arena.dim() == 2 && arena.size(0) == 2 && arena.is_contiguous());
+  int cap = (int)arena.size(1);
+  TORCH_CHECK(n <= cap);
+  hipLaunchKernelGGL(amax_roll_kernel, dim3(cdiv(n, 256)), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     (float*)arena.mutable_data_ptr(), (int)n, cap);
+}
+
 at::Tensor conv2d_fp8_fwd(at::Tensor xq, at::Tensor wq, at::Tensor dq,
+                          c10::optional<at::Tensor> sw,
                           c10::optional<at::Tensor> bias, int64_t stride,
                           int64_t pt, int64_t pb, int64_t pl, int64_t pr,
                           bool reflect, int64_t act, double slope) {
@@ -2036,6 +2124,7 @@ at::Tensor conv2d_fp8_fwd(at::Tensor xq, at::Tensor wq, at::Tensor dq,
   p.w = (const short*)wq.const_data_ptr();
   p.bias = bias.has_value() ? (const short*)bias->const_data_ptr() : nullptr;
   p.dq = (const float*)dq.const_data_ptr();
+  p.dqb = sw.has_value() ? (const float*)sw->const_data_ptr() : nullptr;
   p.y = (short*)y.mutable_data_ptr();
   p.B = xq.size(0); p.H = H; p.W = W; p.Cin = xq.size(3);
   p.OH = OH; p.OW = OW; p.Cout = wq.size(0);

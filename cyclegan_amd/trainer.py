@@ -43,7 +43,8 @@ class CycleGAN:
         self.global_batch_size = args.global_batch_size
         self.compute_dtype = getattr(args, "compute_dtype", None) or (
             torch.bfloat16 if self.device.type == "cuda" else torch.float32)
-        if getattr(args, "fp8", False):
+        self._fp8 = bool(getattr(args, "fp8", False))
+        if self._fp8:
             from .ops import set_fp8_mode
             set_fp8_mode(True)
             self.compute_dtype = torch.bfloat16
@@ -216,6 +217,9 @@ class CycleGAN:
         for g in self.groups.values():
             g.bump_versions()  # invalidate bf16 shadow caches (see flat.py)
         self._refresh_shadows()
+        if getattr(self, "_fp8", False):
+            from .ops import fp8_state
+            fp8_state.roll()  # delayed-scaling amax: prev <- cur
 
         return {
             "loss_G/loss": G_loss.detach(), "loss_G/cycle": G_cycle_loss.detach(),

@@ -139,5 +139,5 @@ sx = torch.tensor([1.0], device=DEV)
 xq = e.quant_fp8(x, sx)
 wq = e.quant_fp8(w, sx)
 dqs = torch.tensor([1.0], device=DEV)
-us8 = timeit(lambda: e.conv2d_fp8_fwd(xq, wq, dqs, None, 1, 1, 1, 1, 1, True, 0, 0.2))
+us8 = timeit(lambda: e.conv2d_fp8_fwd(xq, wq, dqs, None, None, 1, 1, 1, 1, 1, True, 0, 0.2))
 print(f"K3 fp8 fwd   {us8:9.1f} us   {fl8/(us8*1e-6)/1e12:7.1f} TF/s")

@@ -1342,7 +1342,9 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   constexpr int DRPI = 64 / DCH;      // D rows per glds instruction
   constexpr int TDI = (WG_BM * DCH) / 64;   // D glds instructions total
   constexpr int DPI = TDI >= NW ? TDI / NW : 1;  // per wave (maybe idle)
-  constexpr int NFW = WBN >= 32 ? 2 : 1;    // n-fragments per wave
+  // n-fragments per wave: the 4-wave/128-n config packs 4x4 fragments
+  // per wave (32 MFMAs per tr-read set) halving LDS re-read redundancy
+  constexpr int NFW = (NW == 4 && WBN == 128) ? 4 : (WBN >= 32 ? 2 : 1);
   constexpr int NWCW = WBN / (NFW * 16);    // wave-grid columns
   constexpr int KF = (WG_BK / (NW / NWCW)) / 16;  // k-fragments per wave
   extern __shared__ __attribute__((aligned(16))) char wg_smem_raw[];
@@ -2027,15 +2029,26 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
     p.ws = (float*)ws.mutable_data_ptr();
     if (wbn == 128) {
       constexpr size_t SMB = sizeof(WgSmemT<128>);
+      static int wg_nw = []() {
+        const char* e = getenv("CYG_WG_NW");
+        return e ? atoi(e) : 8;
+      }();
       static bool init = []() {
         hipFuncSetAttribute((const void*)(wgrad_glds_kernel<8, 128>),
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            (int)SMB);
+        hipFuncSetAttribute((const void*)(wgrad_glds_kernel<4, 128>),
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             (int)SMB);
         return true;
       }();
       (void)init;
-      hipLaunchKernelGGL((wgrad_glds_kernel<8, 128>), grid, dim3(512), SMB,
-                         stream, p);
+      if (wg_nw == 4)
+        hipLaunchKernelGGL((wgrad_glds_kernel<4, 128>), grid, dim3(256), SMB,
+                           stream, p);
+      else
+        hipLaunchKernelGGL((wgrad_glds_kernel<8, 128>), grid, dim3(512), SMB,
+                           stream, p);
     } else if (wbn == 16) {
       hipLaunchKernelGGL((wgrad_glds_kernel<8, 16>), grid, dim3(512),
                          sizeof(WgSmemT<16>), stream, p);

@@ -313,3 +313,25 @@ def test_conv_512_spatial_shape():
     ref = _conv_ref(x.float().cpu(), w.float().cpu(), None, 1,
                     (3, 3, 3, 3), "reflect")
     check(y, ref, 0.05, "large_spatial")
+
+
+def test_fp8_delayed_scaling_across_steps():
+    """Delayed per-tensor scaling: the first call bootstraps amax; later
+    calls quantize with the rolled previous-step amax. With shrinking
+    activation ranges the stale scale only costs quantization headroom,
+    so every step must stay within fp8 tolerance of the oracle."""
+    from cyclegan_amd.ops import fp8_state
+    from cyclegan_amd.ops import conv as convmod
+    convmod.set_fp8_mode(True)
+    try:
+        w = mk((64, 3, 3, 64), seed=70, scale=0.2).float()
+        for i, s in enumerate((1.0, 0.5, 0.25)):
+            x = mk((2, 16, 16, 64), seed=71 + i) * s
+            y = ops.conv2d(x, w, None, 1, "same", "zeros")
+            ref = _conv_ref(x.float().cpu(), w.cpu(), None, 1,
+                            same_pads(16, 16, 3, 3, 1), "zeros")
+            check(y, ref, 0.15, f"fp8_delayed_{i}")
+            fp8_state.roll()
+            torch.cuda.synchronize()
+    finally:
+        convmod.set_fp8_mode(False)

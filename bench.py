@@ -83,7 +83,8 @@ def main():
     # yet) — every replay still copies the step's input batch in and runs
     # the full forward/backward/optimizer.
     step = gan.train_step
-    if on_gpu and ctx.world_size == 1 and not os.environ.get("CYG_NO_GRAPH"):
+    if on_gpu and not os.environ.get("CYG_NO_GRAPH") and (
+            ctx.world_size == 1 or os.environ.get("CYG_GRAPH_DIST") == "1"):
         from cyclegan_amd.trainer import GraphedStep
         step = GraphedStep(gan, *pool[0])
     sync()

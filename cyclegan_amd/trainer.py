@@ -338,7 +338,12 @@ class GraphedStep:
 
     def __init__(self, gan: CycleGAN, x: torch.Tensor, y: torch.Tensor,
                  warmup: int = 2, preserve_state: bool = False):
-        assert gan.ctx.device.type == "cuda" and gan.ctx.world_size == 1
+        # world_size > 1 capture (RCCL collectives inside the graph) is
+        # proven at 1 rank (profiles/rccl_shakeout.md) but intentionally
+        # opt-in for multi-rank until measured there: CYG_GRAPH_DIST=1.
+        assert gan.ctx.device.type == "cuda" and (
+            gan.ctx.world_size == 1
+            or os.environ.get("CYG_GRAPH_DIST") == "1")
         self.gan = gan
         self.sx = gan._cast(x).clone()
         self.sy = gan._cast(y).clone()

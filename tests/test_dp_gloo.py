@@ -125,3 +125,49 @@ def test_short_final_batch_keeps_replicas_synced(tmp_path):
     r1 = torch.load(os.path.join(str(tmp_path), "short_rank1.pt"))
     assert r0["t"] == r1["t"] == 2
     assert torch.equal(r0["flat_G"], r1["flat_G"])
+
+
+def _worker4(rank, world, port, outdir):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    from cyclegan_amd.parallel import DistContext
+    torch.manual_seed(1234)
+    ctx = DistContext(device=torch.device("cpu"))
+    gan = CycleGAN(_make_args(outdir, 1, 4), ctx)
+    g = torch.Generator().manual_seed(99)
+    x = torch.rand(4, 16, 16, 3, generator=g)
+    y = torch.rand(4, 16, 16, 3, generator=g)
+    gan.train_step(x[rank:rank + 1], y[rank:rank + 1])
+    if rank == 0:
+        torch.save({"flat_G": gan.groups["G"].flat_param.detach().clone()},
+                   os.path.join(outdir, "r4_result.pt"))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_four_rank_matches_single_rank(tmp_path):
+    """4-way DP (the driver runs up to 8): 4 gloo ranks with per-rank
+    batch 1 must equal 1 process with global batch 4."""
+    port = 29523
+    ctxq = mp.get_context("spawn")
+    procs = [ctxq.Process(target=_worker4, args=(r, 4, port, str(tmp_path)))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    got = torch.load(os.path.join(str(tmp_path), "r4_result.pt"),
+                     weights_only=True)
+
+    os.environ.update(RANK="0", WORLD_SIZE="1", LOCAL_RANK="0")
+    from cyclegan_amd.parallel import DistContext
+    torch.manual_seed(1234)
+    ctx = DistContext(device=torch.device("cpu"))
+    gan = CycleGAN(_make_args(str(tmp_path), 4, 4), ctx)
+    g = torch.Generator().manual_seed(99)
+    x = torch.rand(4, 16, 16, 3, generator=g)
+    y = torch.rand(4, 16, 16, 3, generator=g)
+    gan.train_step(x, y)
+    assert torch.allclose(gan.groups["G"].flat_param, got["flat_G"], atol=1e-6)

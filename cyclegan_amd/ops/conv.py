@@ -341,13 +341,16 @@ def conv2d(
         # bf16 packing than from fp8's rate (and skip fp8's quant cost)
         if _head_packable(x, w, stride, pads, pad_mode == "reflect"):
             return _ConvHeadPackedFn.apply(x, w, bias, a, slope)
-        # fp8 only where the 16x16x128 fp8 kernel beats the bf16 glds
-        # kernel (measured): the stride-1 3x3 wide-channel resblock
-        # family. Strided and 4x4 shapes stay bf16 — there the fp8
-        # kernel's per-row decode overhead exceeds the MFMA-rate gain.
+        # fp8 only where the 16x16x128 fp8 kernel + quant beats the bf16
+        # glds kernel (tools/fp8_micro.py on MI355X): the wide-channel
+        # stride-1 3x3 family at batch >= 12 (K3: 999 vs 715 TF/s) and
+        # the 4x4 s1 wide discriminator conv. Elsewhere the per-tensor
+        # quantization overhead cancels the MFMA-rate gain.
         if (_FP8_MODE and x.dtype == torch.bfloat16
-                and stride == 1 and w.shape[1] == 3
-                and w.shape[3] % 16 == 0 and w.shape[3] >= 128):
+                and w.shape[3] % 16 == 0 and w.shape[3] >= 256
+                and stride == 1
+                and ((w.shape[1] == 3 and x.shape[0] >= 12)
+                     or w.shape[1] == 4)):
             return _ConvFp8Fn.apply(x, w, bias, stride, pads,
                                     pad_mode == "reflect", a, slope)
         return _ConvFn.apply(x, w, bias, stride, pads, pad_mode == "reflect", a, slope)

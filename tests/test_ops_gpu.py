@@ -243,7 +243,7 @@ def test_fp8_conv_fwd_vs_oracle():
     from cyclegan_amd.ops import conv as convmod
     convmod.set_fp8_mode(True)
     try:
-        x = mk((2, 16, 16, 256), seed=30)
+        x = mk((12, 16, 16, 256), seed=30)
         w = mk((128, 3, 3, 256), seed=31, scale=0.2).float()
         y = ops.conv2d(x, w, None, 1, (1, 1, 1, 1), "reflect", "relu")
         ref = _conv_ref(x.float().cpu(), w.cpu(), None, 1, (1, 1, 1, 1), "reflect")
@@ -326,7 +326,7 @@ def test_fp8_delayed_scaling_across_steps():
     try:
         w = mk((64, 3, 3, 256), seed=70, scale=0.2).float()
         for i, s in enumerate((1.0, 0.5, 0.25)):
-            x = mk((2, 16, 16, 256), seed=71 + i) * s
+            x = mk((12, 16, 16, 256), seed=71 + i) * s
             y = ops.conv2d(x, w, None, 1, "same", "zeros")
             ref = _conv_ref(x.float().cpu(), w.cpu(), None, 1,
                             same_pads(16, 16, 3, 3, 1), "zeros")

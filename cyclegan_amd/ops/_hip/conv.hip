@@ -1336,15 +1336,19 @@ DEV v4bfx tr16_read(const char* plds) {
       (__attribute__((address_space(3))) v4bfx*)plds);
 }
 
-template <int WBN>
+template <int WBN, int KT = 1>
 struct WgSmemT {
-  short A[2][WG_BM * WG_BK];   // [m][k] swizzled
-  short D[2][WG_BM * WBN];     // [m][n] swizzled
+  short A[2][KT * WG_BM * WG_BK];   // [ktile][m][k] swizzled
+  short D[2][WG_BM * WBN];          // [m][n] swizzled
 };
 
-template <int NW = 4, int WBN = 64>
+// KT = k-tiles per block (2 for the wide resblock shapes): D staging and
+// all per-iteration overhead (advance, loop control, barriers) amortize
+// over KT x the MFMA work; A gather state is per (ktile, instr) slot.
+template <int NW = 4, int WBN = 64, int KT = 1>
 __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   constexpr int API = 16 / NW;        // A glds instructions per wave
+  constexpr int AT = KT * API;        // gather slots per wave
   constexpr int DCH = WBN / 8;        // 16-B chunks per D row
   constexpr int DRPI = 64 / DCH;      // D rows per glds instruction
   constexpr int TDI = (WG_BM * DCH) / 64;   // D glds instructions total
@@ -1355,15 +1359,17 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   constexpr int NWCW = WBN / (NFW * 16);    // wave-grid columns
   constexpr int KF = (WG_BK / (NW / NWCW)) / 16;  // k-fragments per wave
   extern __shared__ __attribute__((aligned(16))) char wg_smem_raw[];
-  auto& sm = *reinterpret_cast<WgSmemT<WBN>*>(wg_smem_raw);
+  auto& sm = *reinterpret_cast<WgSmemT<WBN, KT>*>(wg_smem_raw);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int w = tid >> 6;
   const int wbid = xcd_chunk(blockIdx.x, gridDim.x);
-  const int kt = wbid % p.ktiles;
-  const int nt = (wbid / p.ktiles) % p.ntiles;
-  const int sl = wbid / (p.ktiles * p.ntiles);
+  const int pairs = p.ktiles / KT;    // host guarantees p.ktiles % KT == 0
+  const int ktp = wbid % pairs;
+  const int kt = ktp * KT;            // first k-tile of this block
+  const int nt = (wbid / pairs) % p.ntiles;
+  const int sl = wbid / (pairs * p.ntiles);
   const long k0 = (long)kt * WG_BK;
   const int n0 = nt * WBN;
   const long mstart = sl * p.mchunks_per_slice * WG_BM;
@@ -1374,9 +1380,11 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     // wgrad_reduce sums every chunk, and at::empty memory is dirty
     // (this was a real bug: reused allocator pages leaked garbage into
     // dw at shapes where slices * mchunks_per_slice overshot M)
-    long chunk = (((long)sl * p.ktiles + kt) * p.ntiles + nt) *
-                 ((long)WBN * WG_BK);
-    for (int i = tid; i < WBN * WG_BK; i += NW * 64) p.ws[chunk + i] = 0.f;
+    for (int t = 0; t < KT; ++t) {
+      long chunk = (((long)sl * p.ktiles + kt + t) * p.ntiles + nt) *
+                   ((long)WBN * WG_BK);
+      for (int i = tid; i < WBN * WG_BK; i += NW * 64) p.ws[chunk + i] = 0.f;
+    }
     return;
   }
 
@@ -1387,19 +1395,23 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
 
   // ---- per-lane gather identities ----
   // A: 16 instrs: instr i covers rows i*4..i*4+3; per-wave share = API
-  int a_row[API], a_ci[API], a_dkh[API], a_dkw[API];
-  bool a_kv[API];
+  int a_row[AT], a_ci[AT], a_dkh[AT], a_dkw[AT];
+  bool a_kv[AT];
   #pragma unroll
-  for (int j = 0; j < API; ++j) {
-    int row = (w * API + j) * 4 + (lane >> 4);
-    int cb = ((lane & 15) * 16) ^ AXOR(row);
-    long k = k0 + cb / 2;
-    a_row[j] = row;
-    a_kv[j] = k < p.KTOT;
-    int tap = a_kv[j] ? (int)(k / p.Cin) : 0;
-    a_ci[j] = (int)(k - (long)tap * p.Cin);
-    a_dkh[j] = tap / p.KW;
-    a_dkw[j] = tap - a_dkh[j] * p.KW;
+  for (int t = 0; t < KT; ++t) {
+    #pragma unroll
+    for (int j = 0; j < API; ++j) {
+      int sidx = t * API + j;
+      int row = (w * API + j) * 4 + (lane >> 4);
+      int cb = ((lane & 15) * 16) ^ AXOR(row);
+      long k = k0 + (long)t * WG_BK + cb / 2;
+      a_row[sidx] = row;
+      a_kv[sidx] = k < p.KTOT;
+      int tap = a_kv[sidx] ? (int)(k / p.Cin) : 0;
+      a_ci[sidx] = (int)(k - (long)tap * p.Cin);
+      a_dkh[sidx] = tap / p.KW;
+      a_dkw[sidx] = tap - a_dkh[sidx] * p.KW;
+    }
   }
   // D: instr i covers DRPI rows; per-wave share = DPI
   const bool dw_on = w * DPI < TDI;  // waves beyond TDI stage no D
@@ -1418,9 +1430,9 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   // only at image borders / batch wraps (O(1/OH) of steps).
   const unsigned C2 = (unsigned)p.Cin * 2;
   const unsigned WC2 = (unsigned)p.W * C2;
-  int a_ow[API], a_oh[API], a_b[API], a_ih[API], a_iw[API];
-  unsigned avo[API];
-  bool ainb[API], avalid[API], a_iw_ok[API];
+  int a_ow[AT], a_oh[AT], a_b[AT], a_ih[AT], a_iw[AT];
+  unsigned avo[AT];
+  bool ainb[AT], avalid[AT], a_iw_ok[AT];
   unsigned dvo[DPI];
 
   auto full_a = [&](int j) {
@@ -1438,7 +1450,7 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   };
 
   #pragma unroll
-  for (int j = 0; j < API; ++j) {
+  for (int j = 0; j < AT; ++j) {
     long m = mstart + a_row[j];
     a_ow[j] = (int)(m % p.OW);
     int t = (int)(m / p.OW);
@@ -1467,7 +1479,7 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
       // exec-masked path (1/OH of steps) — this was the 20-mult-per-iter
       // hot spot the compiler was predicating on every iteration.
       #pragma unroll
-      for (int j = 0; j < API; ++j) {
+      for (int j = 0; j < AT; ++j) {
         bool rare = (++a_oh[j] >= p.OH);
         if (!rare) {
           int ih = a_ih[j] + p.stride;
@@ -1489,7 +1501,7 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
       }
     } else {
       #pragma unroll
-      for (int j = 0; j < API; ++j) {
+      for (int j = 0; j < AT; ++j) {
         int ow0 = a_ow[j], oh0 = a_oh[j], b0 = a_b[j];
         a_ow[j] += WG_BM;
         while (a_ow[j] >= p.OW) { a_ow[j] -= p.OW; ++a_oh[j]; }
@@ -1517,11 +1529,17 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
   // straight from the folded avo/dvo registers.
   auto stage = [&](int buf, long ms, bool tail) {
     #pragma unroll
-    for (int j = 0; j < API; ++j) {
-      unsigned vo = (!tail || ms + a_row[j] < p.M) ? avo[j] : 0xFF000000u;
-      __builtin_amdgcn_raw_ptr_buffer_load_lds(
-          rx, (__attribute__((address_space(3))) void*)&sm.A[buf][(w * API + j) * 512],
-          16, vo, 0, 0, 0);
+    for (int t = 0; t < KT; ++t) {
+      #pragma unroll
+      for (int j = 0; j < API; ++j) {
+        int sidx = t * API + j;
+        unsigned vo = (!tail || ms + a_row[sidx] < p.M) ? avo[sidx]
+                                                        : 0xFF000000u;
+        __builtin_amdgcn_raw_ptr_buffer_load_lds(
+            rx, (__attribute__((address_space(3))) void*)
+                &sm.A[buf][t * (WG_BM * WG_BK) + (w * API + j) * 512],
+            16, vo, 0, 0, 0);
+      }
     }
     if (dw_on) {
       #pragma unroll
@@ -1534,7 +1552,7 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     }
   };
 
-  v4f acc[KF][NFW] = {};
+  v4f acc[KT][KF][NFW] = {};
   const int wr = w / NWCW, wc = w % NWCW;
   const int wk0 = wr * (KF * 16), wn0 = wc * (NFW * 16);
   const int fr = lane & 15, fg = lane >> 4;
@@ -1583,14 +1601,6 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
     const char* Db = (const char*)sm.D[cur];
     #pragma unroll
     for (int kk = 0; kk < WG_BM; kk += 32) {
-      // A fragments: MFMA row = weight-k col0+fr, reduce elems = m
-      v8bf a[KF];
-      #pragma unroll
-      for (int kf = 0; kf < KF; ++kf) {
-        v4bfx lo = tr16_read(Ab + kk * 256 + a_base[kf][0]);
-        v4bfx hi = tr16_read(Ab + kk * 256 + a_base[kf][1]);
-        a[kf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
-      }
       v8bf bfr[NFW];
       #pragma unroll
       for (int nf = 0; nf < NFW; ++nf) {
@@ -1598,19 +1608,32 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
         v4bfx hi = tr16_read(Db + kk * (WBN * 2) + d_base[nf][1]);
         bfr[nf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
+      // A fragments: MFMA row = weight-k col0+fr, reduce elems = m
       #pragma unroll
-      for (int kf = 0; kf < KF; ++kf)
+      for (int t = 0; t < KT; ++t) {
+        const char* At = Ab + t * (WG_BM * WG_BK * 2);
+        v8bf a[KF];
         #pragma unroll
-        for (int nf = 0; nf < NFW; ++nf)
-          acc[kf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[kf], bfr[nf], acc[kf][nf], 0, 0, 0);
+        for (int kf = 0; kf < KF; ++kf) {
+          v4bfx lo = tr16_read(At + kk * 256 + a_base[kf][0]);
+          v4bfx hi = tr16_read(At + kk * 256 + a_base[kf][1]);
+          a[kf] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+        }
+        #pragma unroll
+        for (int kf = 0; kf < KF; ++kf)
+          #pragma unroll
+          for (int nf = 0; nf < NFW; ++nf)
+            acc[t][kf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a[kf], bfr[nf], acc[t][kf][nf], 0, 0, 0);
+      }
     }
     __syncthreads();
   }
 
   // slab store (no atomics): chunk layout [WBN][WG_BK], float4 rows
-  {
-    long chunk = (((long)sl * p.ktiles + kt) * p.ntiles + nt) *
+  #pragma unroll
+  for (int t = 0; t < KT; ++t) {
+    long chunk = (((long)sl * p.ktiles + kt + t) * p.ntiles + nt) *
                  ((long)WBN * WG_BK);
     #pragma unroll
     for (int nf = 0; nf < NFW; ++nf) {
@@ -1620,7 +1643,7 @@ __global__ __launch_bounds__(NW * 64) void wgrad_glds_kernel(WgradParams p) {
       for (int kf = 0; kf < KF; ++kf) {
         int kl = wk0 + kf * 16 + fg * 4;
         *(float4*)&p.ws[chunk + (long)nl * WG_BK + kl] =
-            *(const float4*)&acc[kf][nf];
+            *(const float4*)&acc[t][kf][nf];
       }
     }
   }
@@ -2030,16 +2053,36 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
                        p.mchunks_per_slice);
       grid = dim3((long)p.ktiles * p.ntiles * p.slices);
     }
+    static int wg_nw = []() {
+      const char* e = getenv("CYG_WG_NW");
+      return e ? atoi(e) : 8;
+    }();
+    static int wg_kt = []() {
+      const char* e = getenv("CYG_WG_KT");
+      return e ? atoi(e) : 2;
+    }();
+    // 2 k-tiles per block (wide shapes with an even k-tile count):
+    // re-balance the split-M slices for the halved block count
+    bool kt2 = wbn == 128 && wg_nw == 8 && wg_kt == 2 &&
+               (p.ktiles % 2) == 0;
+    if (kt2) {
+      long mchunks2 = (p.M + WG_BM - 1) / WG_BM;
+      int pairs = p.ktiles / 2;
+      int target2 =
+          std::max<long>(1, tgt_blocks / ((long)pairs * p.ntiles));
+      p.slices = (int)std::min<long>(mchunks2, target2);
+      p.mchunks_per_slice = (mchunks2 + p.slices - 1) / p.slices;
+      p.slices = (int)((mchunks2 + p.mchunks_per_slice - 1) /
+                       p.mchunks_per_slice);
+      grid = dim3((long)pairs * p.ntiles * p.slices);
+    }
     auto ws = at::empty({(long)p.slices * p.ktiles * p.ntiles *
                          wbn * WG_BK},
                         x.options().dtype(at::kFloat));
     p.ws = (float*)ws.mutable_data_ptr();
     if (wbn == 128) {
       constexpr size_t SMB = sizeof(WgSmemT<128>);
-      static int wg_nw = []() {
-        const char* e = getenv("CYG_WG_NW");
-        return e ? atoi(e) : 8;
-      }();
+      constexpr size_t SMB2 = sizeof(WgSmemT<128, 2>);
       static bool init = []() {
         hipFuncSetAttribute((const void*)(wgrad_glds_kernel<8, 128>),
                             hipFuncAttributeMaxDynamicSharedMemorySize,
@@ -2047,10 +2090,16 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
         hipFuncSetAttribute((const void*)(wgrad_glds_kernel<4, 128>),
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             (int)SMB);
+        hipFuncSetAttribute((const void*)(wgrad_glds_kernel<8, 128, 2>),
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            (int)SMB2);
         return true;
       }();
       (void)init;
-      if (wg_nw == 4)
+      if (kt2)
+        hipLaunchKernelGGL((wgrad_glds_kernel<8, 128, 2>), grid, dim3(512),
+                           SMB2, stream, p);
+      else if (wg_nw == 4)
         hipLaunchKernelGGL((wgrad_glds_kernel<4, 128>), grid, dim3(256), SMB,
                            stream, p);
       else

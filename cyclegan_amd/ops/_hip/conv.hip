@@ -2058,8 +2058,12 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, int64_t KH, int64_t KW,
       return e ? atoi(e) : 8;
     }();
     static int wg_kt = []() {
+      // measured NEGATIVE (K3 wgrad 65.8 vs 45.2 us; bench 300 vs 326):
+      // the 96 KB LDS footprint drops occupancy to 1 block/CU and the
+      // lost latency hiding dwarfs the amortized staging. CYG_WG_KT=2
+      // to re-enable.
       const char* e = getenv("CYG_WG_KT");
-      return e ? atoi(e) : 2;
+      return e ? atoi(e) : 1;
     }();
     // 2 k-tiles per block (wide shapes with an even k-tile count):
     // re-balance the split-M slices for the halved block count

@@ -79,9 +79,10 @@ def main():
     for i in range(args.warmup):
         gan.train_step(*pool[i % len(pool)])
 
-    # steady-state step as ONE hip graph (N=1; RCCL capture not enabled
-    # yet) — every replay still copies the step's input batch in and runs
-    # the full forward/backward/optimizer.
+    # steady-state step as ONE hip graph (multi-rank capture is opt-in,
+    # CYG_GRAPH_DIST=1 — proven with RCCL at 1 rank) — every replay still
+    # copies the step's input batch in and runs the full
+    # forward/backward/optimizer.
     step = gan.train_step
     if on_gpu and not os.environ.get("CYG_NO_GRAPH") and (
             ctx.world_size == 1 or os.environ.get("CYG_GRAPH_DIST") == "1"):

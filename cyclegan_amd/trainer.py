@@ -15,8 +15,11 @@ on the MI355X execution model:
   backward (reference runs the 4 groups serially, main.py:249-260);
 - G's adversarial gradient flows through the frozen discriminator (the
   ``inputs=`` restriction limits accumulation, not the gradient path);
-  the discriminator update uses re-discriminated detached fakes, matching
-  the reference's recomputed X(fake_x)/Y(fake_y) calls (main.py:239-245);
+  each fake is discriminated ONCE and the node is shared by the
+  adversarial and discriminator losses — gradient-identical to the
+  reference's recomputed X(fake_x)/Y(fake_y) calls (main.py:239-245)
+  under its var_list restriction, asserted by
+  tests/test_trainer.py::test_train_step_grads_match_reference_structure;
 - compute dtype bf16 on GPU (fp32 masters), fp32 on CPU.
 """
 

@@ -83,24 +83,10 @@ def main():
               f"max {e2.max().item():.3f}")
     os.environ.pop("CYG_WG_BLOCKS", None)
 
-    # compare against a second independent oracle: fp64 CPU einsum on a
-    # small slice of m (first 2 samples) is too slow; instead fp64 GPU
-    # unfold for sample 0 only
-    x0 = xrole[:1].double().permute(0, 3, 1, 2)
-    x0 = torch.nn.functional.pad(x0, (pl, pr, pt, pb))
-    dy0 = dyrole[:1].double().permute(0, 3, 1, 2)
-    w64 = torch.zeros(Cout, Cin, KH, KW, device=DEV, dtype=torch.float64,
-                      requires_grad=True)
-    y0 = torch.nn.functional.conv2d(x0, w64, stride=stride)
-    y0.backward(dy0)
-    ref64 = w64.grad.permute(0, 2, 3, 1)
-    # compare fp32 oracle restricted to sample-0 contribution? cannot split;
-    # instead recompute kernel on b=1
-    got0 = E.conv2d_wgrad(xrole[:1].contiguous(), dyrole[:1].contiguous(),
-                          KH, KW, stride, pt, pl, False)
-    e0 = (got0.double() - ref64).abs()
-    b0 = 0.05 * (ref64.abs() + ref64.pow(2).mean().sqrt())
-    print(f"b=1 vs fp64: outliers {(e0 > b0).sum().item()} max {e0.max().item():.3f}")
+    # (a former fp64 b=1 sub-check lived here; its oracle was itself
+    # buggy and produced misleading outlier counts — removed. The fp32
+    # comparison above is the authoritative check; the original bug this
+    # tool diagnosed was idle-slice slab garbage, fixed in conv.hip.)
 
 
 if __name__ == "__main__":

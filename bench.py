@@ -84,10 +84,17 @@ def main():
     # copies the step's input batch in and runs the full
     # forward/backward/optimizer.
     step = gan.train_step
-    if on_gpu and not os.environ.get("CYG_NO_GRAPH") and (
-            ctx.world_size == 1 or os.environ.get("CYG_GRAPH_DIST") == "1"):
-        from cyclegan_amd.trainer import GraphedStep
-        step = GraphedStep(gan, *pool[0])
+    if on_gpu and not os.environ.get("CYG_NO_GRAPH"):
+        if (os.environ.get("CYG_SEGMENTED") != "1"
+                and (ctx.world_size == 1
+                     or os.environ.get("CYG_GRAPH_DIST") == "1")):
+            from cyclegan_amd.trainer import GraphedStep
+            step = GraphedStep(gan, *pool[0])
+        else:
+            # multi-rank: five RCCL-free graphs + eager all-reduces
+            # (CYG_SEGMENTED=1 forces this path for single-rank A/B)
+            from cyclegan_amd.trainer import SegmentedGraphedStep
+            step = SegmentedGraphedStep(gan, *pool[0])
     sync()
 
     t0 = time.perf_counter()

@@ -148,11 +148,11 @@ class CycleGAN:
         self._refresh_shadows()
         return self._zero_losses(self._TRAIN_KEYS)
 
-    def train_step(self, x, y) -> Dict[str, torch.Tensor]:
-        x, y = self._cast(x), self._cast(y)
+    def _forward_losses(self, x, y) -> Dict[str, torch.Tensor]:
+        """The step's forward mega-graph: all 10 loss scalars, grads not
+        yet taken (shared by the eager, fully-graphed, and segmented-graph
+        step drivers)."""
         b = x.shape[0]
-        if b == 0:
-            return self._empty_train_step()
         # batched generator calls: every op is per-sample (convs, per-sample
         # InstanceNorm stats, per-sample losses), so G(cat(x,y)) is
         # numerically identical to G(x), G(y) — fewer, larger kernels.
@@ -185,36 +185,39 @@ class CycleGAN:
         F_cycle_loss = self.cycle_loss(x, cycle_x)
         G_identity_loss = self.identity_loss(y, same_y)
         F_identity_loss = self.identity_loss(x, same_x)
-        G_total = G_loss + G_cycle_loss + G_identity_loss
-        F_total = F_loss + F_cycle_loss + F_identity_loss
+        return {
+            "loss_G/loss": G_loss, "loss_G/cycle": G_cycle_loss,
+            "loss_G/identity": G_identity_loss,
+            "loss_G/total": G_loss + G_cycle_loss + G_identity_loss,
+            "loss_F/loss": F_loss, "loss_F/cycle": F_cycle_loss,
+            "loss_F/identity": F_identity_loss,
+            "loss_F/total": F_loss + F_cycle_loss + F_identity_loss,
+            "loss_X/loss": self.discriminator_loss(self.X(x),
+                                                   discriminate_fake_x),
+            "loss_Y/loss": self.discriminator_loss(self.Y(y),
+                                                   discriminate_fake_y),
+        }
 
-        X_loss = self.discriminator_loss(self.X(x), discriminate_fake_x)
-        Y_loss = self.discriminator_loss(self.Y(y), discriminate_fake_y)
+    _GROUP_LOSS = (("G", "loss_G/total"), ("F", "loss_F/total"),
+                   ("X", "loss_X/loss"), ("Y", "loss_Y/loss"))
 
-        # torch.autograd.grad (not .backward): grads come back as fresh
-        # tensors and land in the flat buffer via ONE batched multi-tensor
-        # copy per group, skipping AccumulateGrad's per-param add into the
-        # .grad views (~300 tiny launches/step) and the flat zero-fill.
-        # retain_graph through X's pass: the shared discriminate_fake_*
-        # subgraphs are traversed by both the generator and discriminator
-        # passes.
-        self.groups["G"].set_grads(
-            torch.autograd.grad(G_total, self.groups["G"].params,
-                                retain_graph=True))
-        self.sync.launch(self.groups["G"].flat_grad)
-        self.groups["F"].set_grads(
-            torch.autograd.grad(F_total, self.groups["F"].params,
-                                retain_graph=True))
-        self.sync.launch(self.groups["F"].flat_grad)
-        self.groups["X"].set_grads(
-            torch.autograd.grad(X_loss, self.groups["X"].params,
-                                retain_graph=True))
-        self.sync.launch(self.groups["X"].flat_grad)
-        self.groups["Y"].set_grads(
-            torch.autograd.grad(Y_loss, self.groups["Y"].params))
-        self.sync.launch(self.groups["Y"].flat_grad)
-        self.sync.wait_all()
+    def _grads_for(self, losses: Dict[str, torch.Tensor], name: str,
+                   retain: bool):
+        """One group's backward pass into its flat grad buffer.
 
+        torch.autograd.grad (not .backward): grads come back as fresh
+        tensors and land in the flat buffer via ONE batched multi-tensor
+        copy per group, skipping AccumulateGrad's per-param add into the
+        .grad views (~300 tiny launches/step) and the flat zero-fill.
+        retain_graph through X's pass: the shared discriminate_fake_*
+        subgraphs are traversed by both the generator and discriminator
+        passes."""
+        key = dict(self._GROUP_LOSS)[name]
+        self.groups[name].set_grads(
+            torch.autograd.grad(losses[key], self.groups[name].params,
+                                retain_graph=retain))
+
+    def _optimize(self):
         for opt in self.optimizers.values():
             opt.step()
         for g in self.groups.values():
@@ -223,6 +226,22 @@ class CycleGAN:
         if getattr(self, "_fp8", False):
             from .ops import fp8_state
             fp8_state.roll()  # delayed-scaling amax: prev <- cur
+
+    def train_step(self, x, y) -> Dict[str, torch.Tensor]:
+        x, y = self._cast(x), self._cast(y)
+        if x.shape[0] == 0:
+            return self._empty_train_step()
+        losses = self._forward_losses(x, y)
+        for i, (name, _) in enumerate(self._GROUP_LOSS):
+            self._grads_for(losses, name, retain=i < 3)
+            self.sync.launch(self.groups[name].flat_grad)
+        self.sync.wait_all()
+        self._optimize()
+        G_loss = losses["loss_G/loss"]; G_cycle_loss = losses["loss_G/cycle"]
+        G_identity_loss = losses["loss_G/identity"]; G_total = losses["loss_G/total"]
+        F_loss = losses["loss_F/loss"]; F_cycle_loss = losses["loss_F/cycle"]
+        F_identity_loss = losses["loss_F/identity"]; F_total = losses["loss_F/total"]
+        X_loss = losses["loss_X/loss"]; Y_loss = losses["loss_Y/loss"]
 
         return {
             "loss_G/loss": G_loss.detach(), "loss_G/cycle": G_cycle_loss.detach(),
@@ -396,6 +415,93 @@ class GraphedStep:
     def call_cloned(self, x, y) -> Dict[str, torch.Tensor]:
         """Replay and return loss scalars detached from the static graph
         outputs (ONE clone kernel), safe to accumulate across steps."""
+        self(x, y)
+        p = self._packed.clone()
+        return {k: p[i] for i, k in enumerate(self.gan._TRAIN_KEYS)}
+
+
+class SegmentedGraphedStep:
+    """Multi-rank graph step: the step is captured as FIVE RCCL-free hip
+    graphs (forward + G-grads, F-grads, X-grads, Y-grads, optimizers)
+    sharing one capture pool, and the four flat-gradient all-reduces are
+    issued EAGERLY between the replays. Multi-rank runs get the captured
+    kernel-launch profile (eager measured ~3.5% slower at N=1) without
+    capturing any collective — RCCL-in-graph capture stays opt-in
+    (CYG_GRAPH_DIST=1, GraphedStep). Collective order matches the eager
+    path exactly (G, F, X, Y), so graphed and eager/empty-slice ranks
+    interoperate."""
+
+    def __init__(self, gan: CycleGAN, x: torch.Tensor, y: torch.Tensor,
+                 warmup: int = 2, preserve_state: bool = False):
+        assert gan.ctx.device.type == "cuda"
+        self.gan = gan
+        self.sx = gan._cast(x).clone()
+        self.sy = gan._cast(y).clone()
+        saved = None
+        if preserve_state:
+            saved = {n: (g.flat_param.clone(), gan.optimizers[n].m.clone(),
+                         gan.optimizers[n].v.clone(), gan.optimizers[n].t)
+                     for n, g in gan.groups.items()}
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(warmup):
+                gan.train_step(self.sx, self.sy)
+        torch.cuda.current_stream().wait_stream(s)
+        for opt in gan.optimizers.values():
+            opt.prepare_graph()
+
+        self.names = [n for n, _ in gan._GROUP_LOSS]
+        self.graphs = []
+        g1 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g1):
+            losses = gan._forward_losses(self.sx, self.sy)
+            gan._grads_for(losses, "G", retain=True)
+        self.graphs.append(g1)
+        pool = g1.pool()
+        for i, name in enumerate(self.names[1:], start=1):
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g, pool=pool):
+                gan._grads_for(losses, name, retain=i < 3)
+            self.graphs.append(g)
+        gopt = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(gopt, pool=pool):
+            for opt in gan.optimizers.values():
+                opt.step()
+            gan._refresh_shadows()
+            if getattr(gan, "_fp8", False):
+                from .ops import fp8_state
+                fp8_state.roll()
+            self._packed = torch.stack([losses[k] for k in gan._TRAIN_KEYS])
+        self.graphs.append(gopt)
+        self.out = {k: losses[k].detach() for k in gan._TRAIN_KEYS}
+
+        if saved is not None:
+            with torch.no_grad():
+                for n, (p, m, v, t) in saved.items():
+                    gan.groups[n].flat_param.copy_(p)
+                    gan.optimizers[n].m.copy_(m)
+                    gan.optimizers[n].v.copy_(v)
+                    gan.optimizers[n].t = t
+                gan._refresh_shadows()
+
+    def __call__(self, x=None, y=None) -> Dict[str, torch.Tensor]:
+        gan = self.gan
+        if x is not None:
+            self.sx.copy_(gan._cast(x))
+            self.sy.copy_(gan._cast(y))
+        for opt in gan.optimizers.values():
+            opt.advance_lr()
+        for i, name in enumerate(self.names):
+            self.graphs[i].replay()
+            gan.sync.launch(gan.groups[name].flat_grad)
+        gan.sync.wait_all()
+        self.graphs[4].replay()
+        for g in gan.groups.values():
+            g.bump_versions()
+        return self.out
+
+    def call_cloned(self, x, y) -> Dict[str, torch.Tensor]:
         self(x, y)
         p = self._packed.clone()
         return {k: p[i] for i, k in enumerate(self.gan._TRAIN_KEYS)}

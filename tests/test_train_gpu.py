@@ -139,3 +139,39 @@ def test_graphed_step_matches_eager(tmp_path):
     # frozen lr_t), which shows up at 1e-2+.
     d = (gan_e.groups["G"].flat_param - gan_g.groups["G"].flat_param)
     assert d.abs().max().item() < 2.5e-3
+
+
+def test_segmented_graphed_step_matches_eager(tmp_path):
+    """Same contract for the multi-rank SegmentedGraphedStep (five
+    RCCL-free graphs sharing one capture pool, all-reduces issued between
+    replays — no-ops at world_size 1): must track the eager trajectory."""
+    from cyclegan_amd.parallel import DistContext
+    from cyclegan_amd.trainer import CycleGAN, SegmentedGraphedStep
+    ctx = DistContext(device=torch.device("cuda", 0))
+    data = []
+    g = torch.Generator().manual_seed(21)
+    for _ in range(4):
+        data.append((torch.rand(2, 64, 64, 3, generator=g) * 2 - 1,
+                     torch.rand(2, 64, 64, 3, generator=g) * 2 - 1))
+
+    torch.manual_seed(5)
+    gan_e = CycleGAN(make_args(tmp_path), ctx)
+    for x, y in data:
+        r_e = gan_e.train_step(x, y)
+    torch.cuda.synchronize()
+
+    torch.manual_seed(5)
+    gan_s = CycleGAN(make_args(tmp_path), ctx)
+    for x, y in data[:2]:
+        gan_s.train_step(x, y)
+    step = SegmentedGraphedStep(gan_s, *data[2], warmup=0)
+    for x, y in data[2:]:
+        r_s = step.call_cloned(x, y)
+    torch.cuda.synchronize()
+
+    assert all(o.t == 4 for o in gan_s.optimizers.values())
+    for k in r_e:
+        a, b = r_e[k].item(), r_s[k].item()
+        assert abs(a - b) <= 0.03 * (abs(b) + 0.03), (k, a, b)
+    d = (gan_e.groups["G"].flat_param - gan_s.groups["G"].flat_param)
+    assert d.abs().max().item() < 2.5e-3

@@ -85,16 +85,18 @@ def main():
     # forward/backward/optimizer.
     step = gan.train_step
     if on_gpu and not os.environ.get("CYG_NO_GRAPH"):
-        if (os.environ.get("CYG_SEGMENTED") != "1"
-                and (ctx.world_size == 1
-                     or os.environ.get("CYG_GRAPH_DIST") == "1")):
-            from cyclegan_amd.trainer import GraphedStep
-            step = GraphedStep(gan, *pool[0])
-        else:
-            # multi-rank: five RCCL-free graphs + eager all-reduces
-            # (CYG_SEGMENTED=1 forces this path for single-rank A/B)
+        if os.environ.get("CYG_SEGMENTED") == "1":
+            # five RCCL-free graphs + eager all-reduces: measured ==
+            # full-graph at N=1 no-dist (324.8 vs 324.3) but == eager
+            # under a communicator (312.9 vs 314.1 at 1-rank force-dist:
+            # the eager AR host overhead, not launch gaps, is the dist
+            # cost) — so it stays opt-in
             from cyclegan_amd.trainer import SegmentedGraphedStep
             step = SegmentedGraphedStep(gan, *pool[0])
+        elif (ctx.world_size == 1
+                or os.environ.get("CYG_GRAPH_DIST") == "1"):
+            from cyclegan_amd.trainer import GraphedStep
+            step = GraphedStep(gan, *pool[0])
     sync()
 
     t0 = time.perf_counter()

@@ -32,20 +32,15 @@ def train(args, pipe, gan, summary, epoch: int):
     results = {}
     # steady-state step as ONE hip graph at N=1 (same path bench.py
     # measures); short final batches and multi-rank training run eager
-    use_graph = (gan.device.type == "cuda"
+    use_graph = (gan.device.type == "cuda" and gan.ctx.world_size == 1
                  and not os.environ.get("CYG_NO_GRAPH"))
     it = DevicePrefetcher(pipe.train_epoch(epoch), gan.device, gan.compute_dtype)
     for x, y in tqdm(it, desc="Train", total=pipe.train_steps,
                      disable=args.verbose == 0 or not gan.ctx.is_main):
         if use_graph and x.shape[0] == args.batch_size:
             if gan.graphed is None:
-                if gan.ctx.world_size == 1:
-                    from cyclegan_amd.trainer import GraphedStep
-                    gan.graphed = GraphedStep(gan, x, y, preserve_state=True)
-                else:
-                    from cyclegan_amd.trainer import SegmentedGraphedStep
-                    gan.graphed = SegmentedGraphedStep(gan, x, y,
-                                                       preserve_state=True)
+                from cyclegan_amd.trainer import GraphedStep
+                gan.graphed = GraphedStep(gan, x, y, preserve_state=True)
             result = gan.graphed.call_cloned(x, y)
         else:
             result = gan.train_step(x, y)
